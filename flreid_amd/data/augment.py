@@ -1,0 +1,92 @@
+"""Tensor-native augmentation (capability parity with
+ref:datasets/image_augmentation.py).
+
+This environment (and the MI355X data path) is tensor-native: datasets store
+CHW float tensors, so augmentations are pure torch ops — no PIL/torchvision.
+The reference's five levels and its (quirky) Normalize-before-Resize order
+are preserved: none/default/rose/sharp/drastic = normalize [+ hflip p=.5 +
+random-erase p∈{.5,.6,.75,.9}] + resize.
+"""
+
+from __future__ import annotations
+
+import random
+from typing import Sequence, Tuple
+
+import torch
+import torch.nn.functional as F
+
+
+class TensorAugment:
+    def __init__(self, size: Tuple[int, int] = (384, 128),
+                 mean: Sequence[float] = (0.485, 0.456, 0.406),
+                 std: Sequence[float] = (0.229, 0.224, 0.225),
+                 hflip_p: float = 0.0, erase_p: float = 0.0,
+                 erase_scale: Tuple[float, float] = (0.02, 0.33),
+                 erase_ratio: Tuple[float, float] = (0.3, 3.3)):
+        self.size = tuple(size)
+        self.mean = torch.tensor(mean).view(-1, 1, 1)
+        self.std = torch.tensor(std).view(-1, 1, 1)
+        self.hflip_p = hflip_p
+        self.erase_p = erase_p
+        self.erase_scale = erase_scale
+        self.erase_ratio = erase_ratio
+
+    def _erase(self, img: torch.Tensor) -> torch.Tensor:
+        c, h, w = img.shape
+        area = h * w
+        for _ in range(10):
+            target = random.uniform(*self.erase_scale) * area
+            ratio = random.uniform(*self.erase_ratio)
+            eh = int(round((target * ratio) ** 0.5))
+            ew = int(round((target / ratio) ** 0.5))
+            if eh < h and ew < w and eh > 0 and ew > 0:
+                y = random.randint(0, h - eh)
+                x = random.randint(0, w - ew)
+                img = img.clone()
+                img[:, y:y + eh, x:x + ew] = torch.randn(c, eh, ew)
+                return img
+        return img
+
+    def __call__(self, img: torch.Tensor) -> torch.Tensor:
+        if not isinstance(img, torch.Tensor):
+            img = torch.as_tensor(img, dtype=torch.float32)
+        img = img.float()
+        img = (img - self.mean) / self.std
+        if self.hflip_p and random.random() < self.hflip_p:
+            img = torch.flip(img, dims=[-1])
+        if self.erase_p and random.random() < self.erase_p:
+            img = self._erase(img)
+        if img.shape[-2:] != self.size:
+            img = F.interpolate(img.unsqueeze(0), size=self.size, mode="bilinear",
+                                align_corners=False).squeeze(0)
+        return img
+
+
+def augmentation_none(size=(384, 128), mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225)):
+    return TensorAugment(size, mean, std)
+
+
+def augmentation_default(size=(384, 128), mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225)):
+    return TensorAugment(size, mean, std, hflip_p=0.5, erase_p=0.5)
+
+
+def augmentation_rose(size=(384, 128), mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225)):
+    return TensorAugment(size, mean, std, hflip_p=0.5, erase_p=0.6)
+
+
+def augmentation_sharp(size=(384, 128), mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225)):
+    return TensorAugment(size, mean, std, hflip_p=0.5, erase_p=0.75)
+
+
+def augmentation_drastic(size=(384, 128), mean=(0.485, 0.456, 0.406), std=(0.229, 0.224, 0.225)):
+    return TensorAugment(size, mean, std, hflip_p=0.5, erase_p=0.9)
+
+
+augmentations = {
+    "none": augmentation_none,
+    "default": augmentation_default,
+    "rose": augmentation_rose,
+    "sharp": augmentation_sharp,
+    "drastic": augmentation_drastic,
+}

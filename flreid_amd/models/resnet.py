@@ -1,0 +1,197 @@
+"""ResNet ReID backbones (capability parity with ref:models/resnet.py).
+
+A from-scratch implementation of the reid-strong-baseline ResNet family:
+  - configurable `last_stride` (stride-1 stage 4 keeps the 16x8 feature map
+    for 128x64 ReID crops — ref:models/resnet.py:182-183)
+  - GAP + optional BNNeck head (BN1d with frozen bias + bias-free classifier,
+    kaiming/small-normal init — ref:models/resnet.py:296-306)
+  - dual-output forward: training returns (cls_score, global_feat), eval
+    returns global_feat (ref:models/resnet.py:312-324)
+
+MI355X notes: parameter/compute dtype is managed by the runtime (bf16 autocast
+with fp32 master weights); convolutions run through MIOpen until the
+hand-written implicit-GEMM HIP kernels (ops/csrc) take over per-layer.
+Submodule names (`base.layer1..4`, `bottleneck`, `classifier`) match the
+reference so `fine_tuning` yaml lists and ckpt schemas transfer unchanged.
+"""
+
+from __future__ import annotations
+
+from typing import List, Type, Union
+
+import torch
+import torch.nn as nn
+
+from flreid_amd.tools.logger import Logger
+from flreid_amd.tools.winit import weights_init_classifier, weights_init_kaiming
+
+_log = Logger("models.resnet")
+
+
+def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+
+
+def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, cin: int, planes: int, stride: int = 1,
+                 downsample: nn.Module = None):
+        super().__init__()
+        self.conv1 = conv3x3(cin, planes, stride)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.relu = nn.ReLU(inplace=True)
+        self.conv2 = conv3x3(planes, planes)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.downsample = downsample
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        return self.relu(out + identity)
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, cin: int, planes: int, stride: int = 1,
+                 downsample: nn.Module = None):
+        super().__init__()
+        self.conv1 = conv1x1(cin, planes)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = conv3x3(planes, planes, stride)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = conv1x1(planes, planes * self.expansion)
+        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        return self.relu(out + identity)
+
+
+class ResNetTrunk(nn.Module):
+    """conv7x7/s2 + maxpool + 4 stages + GAP (ref:models/resnet.py:144-244)."""
+
+    def __init__(self, block: Type[Union[BasicBlock, Bottleneck]],
+                 layers: List[int], last_stride: int = 2):
+        super().__init__()
+        self.inplanes = 64
+        self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.layer1 = self._make_stage(block, 64, layers[0])
+        self.layer2 = self._make_stage(block, 128, layers[1], stride=2)
+        self.layer3 = self._make_stage(block, 256, layers[2], stride=2)
+        self.layer4 = self._make_stage(block, 512, layers[3], stride=last_stride)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self._init_weights()
+
+    def _make_stage(self, block, planes: int, depth: int, stride: int = 1) -> nn.Sequential:
+        downsample = None
+        if stride != 1 or self.inplanes != planes * block.expansion:
+            downsample = nn.Sequential(
+                conv1x1(self.inplanes, planes * block.expansion, stride),
+                nn.BatchNorm2d(planes * block.expansion),
+            )
+        blocks = [block(self.inplanes, planes, stride, downsample)]
+        self.inplanes = planes * block.expansion
+        blocks += [block(self.inplanes, planes) for _ in range(1, depth)]
+        return nn.Sequential(*blocks)
+
+    def _init_weights(self) -> None:
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out", nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.constant_(m.weight, 1.0)
+                nn.init.constant_(m.bias, 0.0)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        x = self.avgpool(x)
+        return torch.flatten(x, 1)
+
+
+_ARCH = {
+    "resnet18": (BasicBlock, [2, 2, 2, 2], 512),
+    "resnet34": (BasicBlock, [3, 4, 6, 3], 512),
+    "resnet50": (Bottleneck, [3, 4, 6, 3], 2048),
+    "resnet101": (Bottleneck, [3, 4, 23, 3], 2048),
+    "resnet152": (Bottleneck, [3, 8, 36, 3], 2048),
+}
+
+
+class ResNetReID(nn.Module):
+    def __init__(self, model_name: str, num_classes: int = 1000,
+                 last_stride: int = 2, neck: str = "no",
+                 pretrained: bool = False, **kwargs):
+        super().__init__()
+        for n, p in kwargs.items():
+            setattr(self, n, p)
+        if model_name not in _ARCH:
+            raise ValueError(f"No model named {model_name}.")
+        block, layers, self.in_planes = _ARCH[model_name]
+        self.model_name = model_name
+        self.num_classes = num_classes
+        self.neck = neck
+        self.base = ResNetTrunk(block, layers, last_stride=last_stride)
+
+        if neck == "no":
+            self.classifier = nn.Linear(self.in_planes, num_classes)
+        elif neck == "bnneck":
+            self.bottleneck = nn.BatchNorm1d(self.in_planes)
+            self.bottleneck.bias.requires_grad_(False)
+            self.classifier = nn.Linear(self.in_planes, num_classes, bias=False)
+            self.bottleneck.apply(weights_init_kaiming)
+            self.classifier.apply(weights_init_classifier)
+        else:
+            raise ValueError(f"Mismatched neck type {neck}.")
+
+        if pretrained:
+            self._try_load_imagenet()
+
+    def _try_load_imagenet(self) -> None:
+        """Optional torch.hub ImageNet weights (ref:models/resnet.py:308-310).
+        This environment has no network; the benchmark path is random-init
+        synthetic (BASELINE.json), so failure is a warning, not an error."""
+        try:
+            from torch.hub import load_state_dict_from_url
+            urls = {
+                "resnet18": "https://download.pytorch.org/models/resnet18-f37072fd.pth",
+                "resnet34": "https://download.pytorch.org/models/resnet34-b627a593.pth",
+                "resnet50": "https://download.pytorch.org/models/resnet50-0676ba61.pth",
+                "resnet101": "https://download.pytorch.org/models/resnet101-63fe2227.pth",
+                "resnet152": "https://download.pytorch.org/models/resnet152-394f9c45.pth",
+            }
+            sd = load_state_dict_from_url(urls[self.model_name], progress=False)
+            sd.pop("fc.weight", None), sd.pop("fc.bias", None)
+            # torchvision naming -> ours: stages/stem names already line up
+            self.base.load_state_dict(sd, strict=False)
+        except Exception as e:  # pragma: no cover
+            _log.warn(f"pretrained weights unavailable ({e}); using random init")
+
+    def forward(self, x: torch.Tensor):
+        global_feat = self.base(x)
+        feat = self.bottleneck(global_feat) if self.neck == "bnneck" else global_feat
+        if self.training:
+            return self.classifier(feat), global_feat
+        return global_feat
+
+
+def resnet18(**kw): return ResNetReID("resnet18", **kw)
+def resnet34(**kw): return ResNetReID("resnet34", **kw)
+def resnet50(**kw): return ResNetReID("resnet50", **kw)
+def resnet101(**kw): return ResNetReID("resnet101", **kw)
+def resnet152(**kw): return ResNetReID("resnet152", **kw)

@@ -1,0 +1,228 @@
+"""Experiment stage — the federated round driver (ref:experiment.py:102-291).
+
+Execution model (the core MI355X redesign):
+  - ONE PROCESS PER GPU (torchrun); rank r owns clients {i : i % W == r}
+    (replaces the reference's VirtualContainer thread-pool device slots,
+    ref:experiment.py:58-99).
+  - The server object is REPLICATED on every rank.  Dispatch states are
+    computed locally from replicated state; uploads are synchronised once per
+    round with a single gather collective (RCCL over xGMI on MI355X), after
+    which `server.calculate()` is a deterministic local computation that
+    yields bitwise-identical server state on every rank (fixed client
+    iteration order, fp32 accumulation).
+  - Online-client sampling is derived from (seed, round), so every rank
+    draws the same set with zero communication (ref:experiment.py:185 used
+    the global RNG).
+
+Single-process mode (world_size == 1) reproduces the reference simulator's
+behaviour on one device — this is the hermetic CPU integration test.
+
+Round structure preserved exactly (ref:experiment.py:183-243):
+dispatch (+ ckpt audit trail `{round}-{server}-{client}.ckpt`) -> train ->
+validate every val_interval (over ALL tasks of each client, for forgetting
+curves) -> upload (+ `{round}-{client}-{server}.ckpt`) -> aggregate.
+"""
+
+from __future__ import annotations
+
+import os
+import random
+from datetime import datetime
+from typing import Any, Dict, List, Optional, Tuple, Union
+
+import torch
+
+from flreid_amd.parallel.comm import FedContext, get_context
+from flreid_amd.runtime.builder import parser_clients, parser_server
+from flreid_amd.runtime.log import ExperimentLog
+from flreid_amd.tools.logger import Logger
+from flreid_amd.tools.utils import clear_cache, same_seeds
+
+
+class ExperimentStage:
+    def __init__(self, common_config: Dict,
+                 exp_configs: Union[Dict, List[Dict]],
+                 ctx: Optional[FedContext] = None):
+        self.common_config = common_config
+        self.exp_configs = [exp_configs] if isinstance(exp_configs, dict) else list(exp_configs)
+        self.logger = Logger("stage")
+        self.ctx = ctx if ctx is not None else get_context()
+        if self.ctx.is_distributed or torch.cuda.is_available():
+            self.device = self.ctx.device
+        else:
+            self.device = common_config.get("device", ["cpu"])[0]
+
+    # ------------------------------------------------------------------ env
+    def __enter__(self):
+        self.check_environment()
+        return self
+
+    def __exit__(self, exc_type, value, trace):
+        return False
+
+    def check_environment(self) -> None:
+        try:
+            torch.tensor([0.0]).to(self.device)
+        except Exception as ex:
+            raise RuntimeError(f"device {self.device} unavailable: {ex}")
+
+        datasets_dir = self.common_config["datasets_dir"]
+        if not datasets_dir.startswith("synthetic:") and not os.path.exists(datasets_dir):
+            raise RuntimeError(f"datasets dir not found: {datasets_dir}")
+
+        ckpt_dir = self.common_config["checkpoints_dir"]
+        if os.path.exists(ckpt_dir) and os.listdir(ckpt_dir):
+            self.logger.warn(f"checkpoint directory {ckpt_dir} is not empty")
+        self.logger.info("Experiment stage build success.")
+
+    # ------------------------------------------------------------------ run
+    def run(self) -> None:
+        for exp_config in self.exp_configs:
+            self.run_experiment(exp_config)
+
+    def run_experiment(self, exp_config: Dict) -> ExperimentLog:
+        same_seeds(exp_config["random_seed"])
+
+        stamp = os.environ.get("FLREID_LOG_STAMP",
+                               datetime.now().strftime("%Y-%m-%d-%H-%M"))
+        log = ExperimentLog(os.path.join(
+            self.common_config["logs_dir"],
+            f"{exp_config['exp_name']}-{stamp}.json"))
+        log.record("config", exp_config)
+        self.logger.info(f"Experiment loading succeed: {exp_config['exp_name']}")
+
+        server = parser_server(exp_config, self.common_config)
+        client_names = [c["client_name"] for c in exp_config["clients"]]
+        owned = [i for i in range(len(client_names))
+                 if self.ctx.owner_of(i) == self.ctx.rank]
+        clients = parser_clients(exp_config, self.common_config, owned_indices=owned)
+        by_name = {c.client_name: c for c in clients}
+
+        # initial validation (round 0) of every owned client (ref:experiment.py:163-173)
+        if exp_config.get("exp_opts", {}).get("initial_validation", True):
+            for client in clients:
+                self._process_val(client, log, 0)
+        log.sync(self.ctx)
+
+        comm_rounds = int(exp_config["exp_opts"]["comm_rounds"])
+        for curr_round in range(1, comm_rounds + 1):
+            if self.ctx.is_rank0():
+                self.logger.info(f"Start communication round: {curr_round:03d}/{comm_rounds:03d}")
+            self.process_one_round(curr_round, server, by_name, client_names,
+                                   exp_config, log)
+        return log
+
+    # ---------------------------------------------------------------- round
+    def _sample_online(self, exp_config: Dict, curr_round: int,
+                       client_names: List[str]) -> List[str]:
+        """Deterministic (seed, round)-keyed draw — identical on all ranks."""
+        k = int(exp_config["exp_opts"]["online_clients"])
+        rng = random.Random(int(exp_config["random_seed"]) * 100003 + curr_round)
+        return rng.sample(client_names, min(k, len(client_names)))
+
+    def process_one_round(self, curr_round: int, server, by_name: Dict[str, Any],
+                          client_names: List[str], exp_config: Dict,
+                          log: ExperimentLog) -> None:
+        online = self._sample_online(exp_config, curr_round, client_names)
+        val_interval = int(exp_config["exp_opts"]["val_interval"])
+        persist_comm = bool(exp_config.get("exp_opts", {}).get("persist_comm_ckpts", True))
+
+        # ---- dispatch (server -> clients); replicated computation ----------
+        for cname in online:
+            if cname not in server.clients:
+                server.register_client(cname)
+                dispatch_state = server.get_dispatch_integrated_state(cname)
+                if dispatch_state is not None and cname in by_name:
+                    by_name[cname].update_by_integrated_state(dispatch_state)
+            else:
+                dispatch_state = server.get_dispatch_incremental_state(cname)
+                if dispatch_state is not None and cname in by_name:
+                    by_name[cname].update_by_incremental_state(dispatch_state)
+            if persist_comm:
+                server.save_state(f"{curr_round}-{server.server_name}-{cname}",
+                                  dispatch_state, True)
+            del dispatch_state
+
+        # ---- local training of owned online clients ------------------------
+        for cname in online:
+            if cname in by_name:
+                self._process_train(by_name[cname], log, curr_round)
+
+        # ---- validation every val_interval rounds --------------------------
+        if val_interval and curr_round % val_interval == 0:
+            for client in by_name.values():
+                self._process_val(client, log, curr_round)
+
+        # ---- upload (clients -> server) ------------------------------------
+        local_uploads: Dict[str, Any] = {}
+        for cname in online:
+            if cname not in by_name:
+                continue
+            client = by_name[cname]
+            state = client.get_incremental_state()
+            if persist_comm:
+                client.save_state(f"{curr_round}-{cname}-{server.server_name}",
+                                  state, True)
+            if state is not None:
+                local_uploads[cname] = _state_to_cpu(state)
+
+        # ---- sync uploads across ranks (ONE gather per round) --------------
+        gathered = self.ctx.all_gather_object(local_uploads)
+        merged: Dict[str, Any] = {}
+        for rank_uploads in gathered:
+            merged.update(rank_uploads)
+        for cname in online:  # deterministic application order
+            if cname in merged:
+                server.set_client_incremental_state(cname, merged[cname])
+
+        # ---- aggregate (replicated, deterministic) -------------------------
+        server.calculate()
+        log.sync(self.ctx)
+
+    # -------------------------------------------------------------- workers
+    @clear_cache
+    def _process_train(self, client, log: ExperimentLog, curr_round: int) -> None:
+        task = client.task_pipeline.next_task()
+        if task["tr_epochs"] == 0:
+            return
+        tr_output = client.train(
+            epochs=task["tr_epochs"],
+            task_name=task["task_name"],
+            tr_loader=task["tr_loader"],
+            val_loader=task["query_loader"],
+            device=self.device,
+        )
+        log.record(f"data.{client.client_name}.{curr_round}.{task['task_name']}",
+                   {"tr_acc": tr_output["accuracy"], "tr_loss": tr_output["loss"]})
+
+    @clear_cache
+    def _process_val(self, client, log: ExperimentLog, curr_round: int) -> None:
+        pipeline = client.task_pipeline
+        for tid in range(len(pipeline.task_list)):
+            task = pipeline.get_task(tid)
+            cmc, mAP, _avg_rep = client.validate(
+                task_name=task["task_name"],
+                query_loader=task["query_loader"],
+                gallery_loader=task["gallery_loaders"],
+                device=self.device,
+            )
+            log.record(f"data.{client.client_name}.{curr_round}.{task['task_name']}", {
+                "val_rank_1": float(cmc[0]),
+                "val_rank_3": float(cmc[2]) if len(cmc) > 2 else float(cmc[-1]),
+                "val_rank_5": float(cmc[4]) if len(cmc) > 4 else float(cmc[-1]),
+                "val_rank_10": float(cmc[9]) if len(cmc) > 9 else float(cmc[-1]),
+                "val_map": mAP,
+            })
+
+
+def _state_to_cpu(state: Any) -> Any:
+    """Detach+cpu a nested state for the object gather path."""
+    if torch.is_tensor(state):
+        return state.detach().cpu()
+    if isinstance(state, dict):
+        return {k: _state_to_cpu(v) for k, v in state.items()}
+    if isinstance(state, list):
+        return [_state_to_cpu(v) for v in state]
+    if isinstance(state, tuple):
+        return tuple(_state_to_cpu(v) for v in state)
+    return state
