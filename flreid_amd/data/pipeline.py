@@ -57,7 +57,8 @@ class ReIDTaskPipeline:
             so = self.synthetic_opts
             per_id = {"train": so["train"], "query": so["query"], "gallery": so["gallery"]}[split]
             return SyntheticReIDDataset(task, split, so["ids"], per_id,
-                                        so["shape"], so["base"], transform)
+                                        so["shape"], so["base"], transform,
+                                        idspace=so["idspace"])
         return ReIDImageDataset(os.path.join(self.datasets_dir, task, split), transform)
 
     def get_task(self, idx: int = -1) -> Dict:

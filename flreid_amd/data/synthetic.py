@@ -35,7 +35,8 @@ def parse_synthetic_dir(datasets_dir: str) -> Optional[Dict]:
     if not datasets_dir.startswith("synthetic:"):
         return None
     spec = datasets_dir.split("//", 1)[-1]
-    opts = {"ids": 32, "train": 4, "query": 2, "gallery": 4, "hw": "128x64", "base": 0}
+    opts = {"ids": 32, "train": 4, "query": 2, "gallery": 4, "hw": "128x64",
+            "base": 0, "idspace": 4096}
     if spec:
         for kv in spec.split(","):
             if not kv:
@@ -57,7 +58,8 @@ class SyntheticReIDDataset(Dataset):
 
     def __init__(self, task_name: str, split: str, n_ids: int,
                  imgs_per_id: int, shape: Tuple[int, int, int] = (3, 128, 64),
-                 id_base: int = 0, transform: Callable = None):
+                 id_base: int = 0, transform: Callable = None,
+                 idspace: int = 4096):
         super().__init__()
         self.task_name = task_name
         self.split = split
@@ -65,7 +67,11 @@ class SyntheticReIDDataset(Dataset):
         self.imgs_per_id = imgs_per_id
         self.shape = shape
         self.transform = transform
-        task_stride = _seed_from(task_name, "ids") % 4096
+        # person ids stay inside [id_base, id_base + idspace) so they always
+        # fit the classifier head (num_classes >= id_base + idspace)
+        slots = max(1, idspace // max(1, n_ids))
+        task_stride = (_seed_from(task_name, "ids") % slots) * n_ids
+        task_stride = min(task_stride, max(0, idspace - n_ids))
         self.classes = [id_base + task_stride + i for i in range(n_ids)]
 
     @property

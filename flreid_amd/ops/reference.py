@@ -220,7 +220,10 @@ def cmc_map(query_features: torch.Tensor, query_labels: torch.Tensor,
         junk = torch.zeros_like(same_id)
         good = same_id
 
-    order = torch.argsort(sims, dim=1, descending=True, stable=True)  # [Q, G]
+    # np.argsort(sim)[::-1] semantics: ascending stable sort then reverse —
+    # on ties the HIGHER gallery index ranks first (ref:tools/evaluate.py:129)
+    order = torch.flip(torch.argsort(sims, dim=1, descending=False, stable=True),
+                       dims=[1])                                      # [Q, G]
     good_sorted = good.gather(1, order)
     valid_sorted = (~junk).gather(1, order)
 

@@ -144,8 +144,14 @@ class ExperimentStage:
             del dispatch_state
 
         # ---- local training of owned online clients ------------------------
+        # per-(client, round) seeding keeps a client's training stream
+        # independent of which rank hosts it and of sibling clients' order,
+        # so W=1 and W=N runs are bitwise comparable (tests/test_comm_gloo.py)
+        base_seed = int(exp_config["random_seed"])
         for cname in online:
             if cname in by_name:
+                same_seeds((base_seed * 1000003 + curr_round * 1009 +
+                            client_names.index(cname)) % (2 ** 31))
                 self._process_train(by_name[cname], log, curr_round)
 
         # ---- validation every val_interval rounds --------------------------

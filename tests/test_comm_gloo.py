@@ -1,0 +1,180 @@
+"""Multi-process federation tests on CPU/gloo (world_size=2).
+
+The load-bearing test: a 2-rank distributed FedAvg run must produce the SAME
+final server model as the single-process simulator run — the collectives are
+semantically transparent (SURVEY.md §5.8 correctness crux).
+"""
+
+import json
+import os
+import pickle
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _dist_env(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["FLREID_LOG_STAMP"] = "test"
+
+
+def _worker_allreduce(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        t = {"w": torch.full((4,), float(rank + 1))}
+        # rank0 weight .25, rank1 weight .75
+        out = ctx.weighted_allreduce(t, 0.25 if rank == 0 else 0.75)
+        expected = 1.0 * 0.25 + 2.0 * 0.75
+        assert torch.allclose(out["w"], torch.full((4,), expected))
+
+        gathered = ctx.all_gather_object({"rank": rank})
+        assert [g["rank"] for g in gathered] == [0, 1]
+
+        flat = ctx.all_gather_flat(torch.arange(3, dtype=torch.float32) + rank * 10)
+        assert flat.shape == (2, 3)
+        assert torch.allclose(flat[1], torch.tensor([10.0, 11.0, 12.0]))
+    finally:
+        destroy_context()
+
+
+def test_collectives_gloo(tmp_path):
+    port = _free_port()
+    mp.spawn(_worker_allreduce, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+
+
+def _run_experiment(tmpdir, tag):
+    """Build config dicts locally (spawned workers import fresh)."""
+    common = {
+        "datasets_dir": "synthetic://ids=4,train=3,query=2,gallery=3,hw=32x16,idspace=48",
+        "checkpoints_dir": os.path.join(tmpdir, f"ckpts-{tag}"),
+        "logs_dir": os.path.join(tmpdir, f"logs-{tag}"),
+        "parallel": 1, "device": ["cpu"], "defaults": {},
+    }
+    exp = {
+        "exp_name": "dist", "exp_method": "fedavg", "random_seed": 11,
+        "exp_opts": {"comm_rounds": 2, "val_interval": 0, "online_clients": 2,
+                     "initial_validation": False},
+        "model_opts": {"name": "resnet18", "num_classes": 64, "last_stride": 1,
+                       "neck": "bnneck",
+                       "fine_tuning": ["base.layer4", "classifier"]},
+        "criterion_opts": {"name": "cross_entropy", "num_classes": 64,
+                           "epsilon": 0.1},
+        "optimizer_opts": {"name": "adam", "lr": 1e-3, "weight_decay": 1e-5},
+        "scheduler_opts": {"name": "step_lr", "step_size": 5},
+        "task_opts": {"sustain_rounds": 1, "train_epochs": 1,
+                      "augment_opts": {"level": "none", "img_size": [32, 16],
+                                       "norm_mean": [0.485, 0.456, 0.406],
+                                       "norm_std": [0.229, 0.224, 0.225]},
+                      "loader_opts": {"batch_size": 8, "num_workers": 0,
+                                      "pin_memory": False,
+                                      "persistent_workers": False,
+                                      "multiprocessing_context": None}},
+        "server": {"server_name": "server"},
+        "clients": [
+            {"client_name": "client-0", "tasks": ["task-0-0"]},
+            {"client_name": "client-1", "tasks": ["task-1-0"]},
+        ],
+    }
+    return common, exp
+
+
+def _final_server_state(common, exp):
+    from flreid_amd.parallel.comm import get_context
+    from flreid_amd.runtime.builder import parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+
+    stage = ExperimentStage(common, [exp])
+    stage.run_experiment(exp)
+    # re-derive the server's final trainable params by rebuilding + reloading?
+    # no — grab from the stage-owned server via a fresh run instead:
+    return None
+
+
+def _worker_dist_e2e(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    from flreid_amd.runtime.builder import parser_clients, parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+    from flreid_amd.tools.utils import same_seeds
+
+    ctx = init_context(device="cpu")
+    try:
+        common, exp = _run_experiment(tmpdir, "dist")
+        stage = ExperimentStage(common, [exp], ctx=ctx)
+
+        # run (mirrors run_experiment but keeps the server for inspection)
+        same_seeds(exp["random_seed"])
+        from flreid_amd.runtime.log import ExperimentLog
+        log = ExperimentLog(os.path.join(common["logs_dir"], "dist-test.json"))
+        server = parser_server(exp, common)
+        client_names = [c["client_name"] for c in exp["clients"]]
+        owned = [i for i in range(len(client_names)) if ctx.owner_of(i) == ctx.rank]
+        clients = parser_clients(exp, common, owned_indices=owned)
+        by_name = {c.client_name: c for c in clients}
+        assert len(clients) == 1   # one client per rank
+
+        for r in (1, 2):
+            stage.process_one_round(r, server, by_name, client_names, exp, log)
+
+        if rank == 0:
+            state = {n: p.detach().clone() for n, p in
+                     server.model.named_parameters() if p.requires_grad}
+            with open(os.path.join(tmpdir, "dist_server_state.pkl"), "wb") as f:
+                pickle.dump(state, f)
+    finally:
+        destroy_context()
+
+
+@pytest.mark.timeout(600)
+def test_distributed_equals_single_process(tmp_path):
+    tmpdir = str(tmp_path)
+
+    # --- single process run -------------------------------------------------
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    from flreid_amd.parallel.comm import FedContext
+    from flreid_amd.runtime.builder import parser_clients, parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+    from flreid_amd.runtime.log import ExperimentLog
+    from flreid_amd.tools.utils import same_seeds
+
+    common, exp = _run_experiment(tmpdir, "single")
+    ctx = FedContext()
+    stage = ExperimentStage(common, [exp], ctx=ctx)
+    same_seeds(exp["random_seed"])
+    log = ExperimentLog(os.path.join(common["logs_dir"], "single-test.json"))
+    server = parser_server(exp, common)
+    client_names = [c["client_name"] for c in exp["clients"]]
+    clients = parser_clients(exp, common)
+    by_name = {c.client_name: c for c in clients}
+    for r in (1, 2):
+        stage.process_one_round(r, server, by_name, client_names, exp, log)
+    single_state = {n: p.detach().clone() for n, p in
+                    server.model.named_parameters() if p.requires_grad}
+
+    # --- 2-rank distributed run --------------------------------------------
+    port = _free_port()
+    mp.spawn(_worker_dist_e2e, args=(2, port, tmpdir), nprocs=2, join=True)
+
+    with open(os.path.join(tmpdir, "dist_server_state.pkl"), "rb") as f:
+        dist_state = pickle.load(f)
+
+    assert set(single_state) == set(dist_state)
+    for n in single_state:
+        assert torch.allclose(single_state[n], dist_state[n], atol=1e-6), n

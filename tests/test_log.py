@@ -1,0 +1,31 @@
+"""ExperimentLog semantics (ref:experiment.py:16-55)."""
+
+import json
+
+from flreid_amd.runtime.log import ExperimentLog
+
+
+def test_dotted_nested_keys(tmp_path):
+    log = ExperimentLog(str(tmp_path / "log.json"))
+    log.record("data.client-0.3.task-0-1", {"tr_acc": 0.5})
+    log.record("data.client-0.3.task-0-1", {"tr_loss": 1.0})  # dict merge
+    assert log.records["data"]["client-0"]["3"]["task-0-1"] == {
+        "tr_acc": 0.5, "tr_loss": 1.0}
+
+
+def test_list_append_and_scalar_replace(tmp_path):
+    log = ExperimentLog(str(tmp_path / "log.json"))
+    log.record("a", [1])
+    log.record("a", 2)   # appended to existing list
+    assert log.records["a"] == [1, 2]
+    log.record("b", 1)
+    log.record("b", 3)   # scalar replaced
+    assert log.records["b"] == 3
+
+
+def test_flush_writes_json(tmp_path):
+    path = tmp_path / "x" / "log.json"
+    log = ExperimentLog(str(path))
+    log.record("config", {"exp_name": "t"})
+    log.flush()
+    assert json.loads(path.read_text())["config"]["exp_name"] == "t"
