@@ -1,0 +1,189 @@
+"""FedSTIL additive-decomposition layers (ref:methods/fedstil.py:24-225).
+
+θ = global_weight_atten ⊙ global_weight + adaptive_weight, with the attention
+vector broadcast over the LAST weight dimension
+(ref:methods/fedstil.py:66-67,84-92).  Only `adaptive_weight` (and bias)
+train; `global_weight` is the federated state and `global_weight_atten` is a
+fixed mixing vector (it becomes a learnable stacked attention in the
+fedstil-atten variant — ref:methods/fedstil_atten.py).
+
+`init_training_weights()` (no args) is the dispatch-time re-initialisation:
+atten resets to `atten_default` and adaptive_weight to (1−atten)·W_glob, so
+the composed weight right after dispatch equals the dispatched global weight
+(ref:methods/fedstil.py:53-82, applied at ref:methods/fedstil.py:879-911).
+`initial_*` snapshots anchor the round's L1 drift regulariser
+(ref:methods/fedstil.py:639-644).
+
+On GPU the composition runs through the fused HIP compose kernel
+(flreid_amd.ops.adaptive_compose) so θ is produced in one pass; the backward
+for adaptive_weight is identity and atten (when learnable) reduces over all
+dims but the last.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+from torch.nn import Parameter
+
+from flreid_amd import ops
+
+
+class AdaptiveBase(nn.Module):
+    """Holds the decomposed parameter set; subclasses implement forward."""
+
+    def __init__(self, global_weight: torch.Tensor,
+                 global_weight_atten: Optional[torch.Tensor] = None,
+                 adaptive_weight: Optional[torch.Tensor] = None,
+                 adaptive_bias: Optional[torch.Tensor] = None,
+                 atten_default: float = 0.80, atten_trainable: bool = False,
+                 **kwargs):
+        super().__init__()
+        self.atten_default = atten_default
+        self.atten_trainable = atten_trainable
+        self.global_weight = Parameter(torch.empty(0), requires_grad=False)
+        self.global_weight_atten = Parameter(torch.empty(0), requires_grad=False)
+        self.adaptive_weight = Parameter(torch.empty(0))
+        self.adaptive_bias = Parameter(torch.empty(0)) if adaptive_bias is not None else None
+        self.initial_global_weight_atten = Parameter(torch.empty(0), requires_grad=False)
+        self.initial_adaptive_weight = Parameter(torch.empty(0), requires_grad=False)
+        self.init_training_weights(global_weight, global_weight_atten,
+                                   adaptive_weight, adaptive_bias)
+
+    @torch.no_grad()
+    def init_training_weights(self, global_weight=None, global_weight_atten=None,
+                              adaptive_weight=None, adaptive_bias=None) -> None:
+        if global_weight is None:
+            global_weight = self.global_weight.data
+        self.global_weight.data = global_weight.detach().clone()
+        self.global_weight.requires_grad = False
+
+        if global_weight_atten is None:
+            global_weight_atten = torch.ones(
+                self.global_weight.data.shape[-1],
+                device=self.global_weight.device) * self.atten_default
+        self.global_weight_atten.data = global_weight_atten.detach().clone()
+        self.initial_global_weight_atten.data = global_weight_atten.detach().clone()
+        self.global_weight_atten.requires_grad = self.atten_trainable
+
+        if adaptive_weight is None:
+            adaptive_weight = (1.0 - self.global_weight_atten.data) * self.global_weight.data
+        self.adaptive_weight.data = adaptive_weight.detach().clone()
+        self.initial_adaptive_weight.data = adaptive_weight.detach().clone()
+        self.adaptive_weight.requires_grad = True
+
+        if self.adaptive_bias is not None and adaptive_bias is not None:
+            self.adaptive_bias.data = adaptive_bias.detach().clone()
+            self.adaptive_bias.requires_grad = True
+
+    def composed_weight(self) -> torch.Tensor:
+        return ops.adaptive_compose(self.global_weight, self.global_weight_atten,
+                                    self.adaptive_weight)
+
+    def drift_pairs(self):
+        """(current, round-start) pairs for the L1 drift regulariser."""
+        return [(self.global_weight_atten, self.initial_global_weight_atten),
+                (self.adaptive_weight, self.initial_adaptive_weight)]
+
+
+class AdaptiveLinear(AdaptiveBase):
+    def forward(self, data: torch.Tensor) -> torch.Tensor:
+        return F.linear(data, self.composed_weight(), self.adaptive_bias)
+
+
+class AdaptiveConv2d(AdaptiveBase):
+    def __init__(self, global_weight, stride=1, padding=0, **kwargs):
+        super().__init__(global_weight, **kwargs)
+        self.stride = stride
+        self.padding = padding
+
+    def forward(self, data: torch.Tensor) -> torch.Tensor:
+        return F.conv2d(data, self.composed_weight(), self.adaptive_bias,
+                        stride=self.stride, padding=self.padding)
+
+
+class AdaptiveBatchNorm2d(AdaptiveBase):
+    """Completeness parity with ref:methods/fedstil.py:131-198 (the reference
+    ships this but keeps BatchNorm out of its transform LUT)."""
+
+    def __init__(self, global_weight, running_mean=None, running_var=None,
+                 num_batches_tracked=None, track_running_stats=False,
+                 momentum=0.1, eps=1e-5, **kwargs):
+        super().__init__(global_weight, **kwargs)
+        self.register_buffer("running_mean", running_mean)
+        self.register_buffer("running_var", running_var)
+        self.register_buffer("num_batches_tracked", num_batches_tracked)
+        self.track_running_stats = track_running_stats
+        self.momentum = momentum
+        self.eps = eps
+
+    def forward(self, data: torch.Tensor) -> torch.Tensor:
+        training = self.training or (self.running_mean is None and self.running_var is None)
+        use_stats = not self.training or self.track_running_stats
+        return F.batch_norm(
+            data,
+            self.running_mean if use_stats else None,
+            self.running_var if use_stats else None,
+            self.composed_weight(), self.adaptive_bias,
+            training, self.momentum or 0.0, self.eps)
+
+
+class AdaptiveLayerNorm(AdaptiveBase):
+    def __init__(self, global_weight, normalized_shape: Tuple[int, ...] = None,
+                 eps: float = 1e-5, **kwargs):
+        super().__init__(global_weight, **kwargs)
+        self.normalized_shape = normalized_shape or tuple(global_weight.shape)
+        self.eps = eps
+
+    def forward(self, data: torch.Tensor) -> torch.Tensor:
+        return F.layer_norm(data, self.normalized_shape, self.composed_weight(),
+                            self.adaptive_bias, self.eps)
+
+
+ADAPTIVE_TYPES = (AdaptiveLinear, AdaptiveConv2d, AdaptiveBatchNorm2d, AdaptiveLayerNorm)
+
+
+def convert_to_adaptive(net: nn.Module, atten_default: float = 0.80,
+                        atten_trainable: bool = False) -> int:
+    """Rewrite every fully-trainable Linear/Conv2d leaf into its adaptive
+    counterpart (ref:methods/fedstil.py:290-347; BN/LN stay untouched like the
+    reference's LUT).  Returns the number of converted leaves."""
+    count = 0
+    for name, module in list(net.named_modules()):
+        if isinstance(module, (nn.Linear, nn.Conv2d)):
+            if not all(p.requires_grad for p in module.parameters()):
+                continue
+            if isinstance(module, nn.Linear):
+                new = AdaptiveLinear(
+                    global_weight=module.weight, adaptive_bias=module.bias,
+                    atten_default=atten_default, atten_trainable=atten_trainable)
+            else:
+                new = AdaptiveConv2d(
+                    global_weight=module.weight, adaptive_bias=module.bias,
+                    atten_default=atten_default, atten_trainable=atten_trainable,
+                    stride=module.stride, padding=module.padding)
+            parent = net
+            parts = name.split(".")
+            for p in parts[:-1]:
+                parent = getattr(parent, p)
+            setattr(parent, parts[-1], new)
+            count += 1
+    return count
+
+
+def adaptive_leaves(net: nn.Module):
+    """[(qualname, module)] of adaptive leaves, in module order."""
+    return [(n, m) for n, m in net.named_modules() if isinstance(m, ADAPTIVE_TYPES)]
+
+
+def non_adaptive_leaves(net: nn.Module):
+    """Leaves that are NOT adaptive (the 'pre-trained' frozen set,
+    ref:methods/fedstil.py:421-433)."""
+    out = []
+    for n, m in net.named_modules():
+        if len(list(m.children())) == 0 and not isinstance(m, ADAPTIVE_TYPES) and n:
+            out.append((n, m))
+    return out

@@ -182,12 +182,49 @@ class ResNetReID(nn.Module):
         except Exception as e:  # pragma: no cover
             _log.warn(f"pretrained weights unavailable ({e}); using random init")
 
-    def forward(self, x: torch.Tensor):
-        global_feat = self.base(x)
+    # ---- staged execution --------------------------------------------------
+    # FedSTIL trains only the "head" (everything from the first adaptive layer
+    # on) on cached prototype features.  The reference located that split with
+    # a torch.fx graph surgery (ref:methods/fedstil.py:258-288 +
+    # ref:tools/utils.py:139-182); this framework owns its model definitions,
+    # so the split is an explicit stage list — no tracer, no graph rewrite.
+    STAGES = ("stem", "layer1", "layer2", "layer3", "layer4", "head")
+
+    def stage_of(self, module_path: str) -> int:
+        """Stage index that contains the (dotted) submodule path."""
+        if module_path.startswith("base.layer"):
+            return self.STAGES.index(module_path.split(".")[1])
+        if module_path.startswith("base."):
+            return 0
+        return len(self.STAGES) - 1          # bottleneck / classifier
+
+    def run_stages(self, x: torch.Tensor, start: int = 0, tap: int = None):
+        """Run stages [start, end); optionally capture the INPUT of stage
+        `tap` (the prototype feature FedSTIL caches).  Returns (output,
+        tap_value) with output matching forward()'s train/eval convention."""
+        tap_value = None
+        b = self.base
+        stages = [
+            lambda t: b.maxpool(b.relu(b.bn1(b.conv1(t)))),
+            b.layer1, b.layer2, b.layer3, b.layer4,
+        ]
+        for idx in range(start, 5):
+            if tap == idx:
+                tap_value = x
+            x = stages[idx](x)
+        if tap == 5:
+            tap_value = x
+        # the head stage consumes the (spatial) layer4 output: pool -> neck ->
+        # classifier; a head-only invocation passes the cached spatial tap in
+        global_feat = torch.flatten(b.avgpool(x), 1)
         feat = self.bottleneck(global_feat) if self.neck == "bnneck" else global_feat
         if self.training:
-            return self.classifier(feat), global_feat
-        return global_feat
+            return (self.classifier(feat), global_feat), tap_value
+        return global_feat, tap_value
+
+    def forward(self, x: torch.Tensor):
+        out, _ = self.run_stages(x, 0)
+        return out
 
 
 def resnet18(**kw): return ResNetReID("resnet18", **kw)
