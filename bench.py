@@ -1,0 +1,183 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: FedSTIL ResNet-50 federated round throughput on MI355X.
+
+Measures the BASELINE.json metric — images/sec (whole node) per federated
+round, FedSTIL ResNet-50, one simulated edge client per GPU, bf16 compute,
+synthetic data, random-init weights.  One bench step == one communication
+round: prototype-capture forward over the task loader (frozen backbone) +
+head training epoch on prototype features + upload / aggregate / dispatch
+(RCCL collectives when N > 1).
+
+Launch (the driver's contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Weak scaling: per-GPU work is fixed (one client per rank, same task size);
+the whole-job value aggregates over all ranks.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
+if REPO_ROOT not in sys.path:
+    sys.path.insert(0, REPO_ROOT)
+
+# benchmark mode: no ckpt audit trail in the timed loop
+os.environ.setdefault("FLREID_DISABLE_CKPT", "1")
+
+import torch  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", type=str, default="resnet50")
+    p.add_argument("--method", type=str, default="fedstil")
+    p.add_argument("--batch", type=int, default=64)
+    p.add_argument("--img", type=str, default="128x64")
+    p.add_argument("--ids", type=int, default=64,
+                   help="identities per synthetic task")
+    p.add_argument("--imgs-per-id", type=int, default=8,
+                   help="train images per identity per task")
+    p.add_argument("--num-classes", type=int, default=8000)
+    p.add_argument("--cpu", action="store_true", help="debug on CPU")
+    return p.parse_args()
+
+
+def build_configs(args, world_size, rank):
+    h, w = (int(x) for x in args.img.split("x"))
+    n_clients = max(args.gpus, world_size)
+    common = {
+        "datasets_dir": (f"synthetic://ids={args.ids},train={args.imgs_per_id},"
+                         f"query=2,gallery=4,hw={h}x{w},"
+                         f"idspace={min(4096, args.num_classes)}"),
+        "checkpoints_dir": "./gpurun_out/bench_ckpts/",
+        "logs_dir": "./gpurun_out/bench_logs/",
+        "parallel": 1,
+        "device": ["cpu" if args.cpu else "cuda:0"],
+        "defaults": {},
+    }
+    exp = {
+        "exp_name": "bench", "exp_method": args.method, "random_seed": 42,
+        "exp_opts": {"comm_rounds": args.steps + args.warmup,
+                     "val_interval": 0, "online_clients": n_clients,
+                     "initial_validation": False,
+                     "persist_comm_ckpts": False},
+        "model_opts": {"name": args.model, "num_classes": args.num_classes,
+                       "last_stride": 1, "neck": "bnneck",
+                       "atten_default": 0.9, "lambda_l1": 1e-4,
+                       "lambda_k": 2000,
+                       "fine_tuning": ["base.layer4", "classifier"]},
+        "criterion_opts": {"name": "cross_entropy",
+                           "num_classes": args.num_classes, "epsilon": 0.1},
+        "optimizer_opts": {"name": "adam", "lr": 1e-3, "weight_decay": 1e-5},
+        "scheduler_opts": {"name": "step_lr", "step_size": 5},
+        "task_opts": {"sustain_rounds": 10 ** 9, "train_epochs": 1,
+                      "augment_opts": {"level": "default", "img_size": [h, w],
+                                       "norm_mean": [0.485, 0.456, 0.406],
+                                       "norm_std": [0.229, 0.224, 0.225]},
+                      "loader_opts": {"batch_size": args.batch,
+                                      "num_workers": 0, "pin_memory": False,
+                                      "persistent_workers": False,
+                                      "multiprocessing_context": None}},
+        "server": {"server_name": "server", "distance_calculate_step": 10,
+                   "distance_calculate_decay": 0.8},
+        "clients": [{"client_name": f"client-{i}",
+                     "tasks": [f"task-{i}-0"]} for i in range(n_clients)],
+    }
+    return common, exp
+
+
+def main():
+    args = parse_args()
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    from flreid_amd.runtime.builder import parser_clients, parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+    from flreid_amd.runtime.log import ExperimentLog
+    from flreid_amd.tools.utils import same_seeds
+
+    ctx = init_context(device="cpu" if args.cpu else None)
+    device_is_cuda = ctx.device.startswith("cuda")
+    if device_is_cuda:
+        from flreid_amd import ops
+        if not ops.extension_available():
+            raise RuntimeError(
+                "flreid HIP extension not built — run "
+                "`python -m flreid_amd.ops.build` before benchmarking")
+
+    common, exp = build_configs(args, ctx.world_size, ctx.rank)
+    os.makedirs(common["logs_dir"], exist_ok=True)
+
+    same_seeds(exp["random_seed"])
+    stage = ExperimentStage(common, [exp], ctx=ctx)
+    log = ExperimentLog(os.path.join(common["logs_dir"], "bench.json"))
+
+    server = parser_server(exp, common)
+    client_names = [c["client_name"] for c in exp["clients"]]
+    owned = [i for i in range(len(client_names))
+             if ctx.owner_of(i) == ctx.rank]
+    clients = parser_clients(exp, common, owned_indices=owned)
+    by_name = {c.client_name: c for c in clients}
+
+    def one_round(r):
+        stage.process_one_round(r, server, by_name, client_names, exp, log)
+
+    # images processed per round per client = the task's train split size
+    imgs_per_client = args.ids * args.imgs_per_id
+    n_clients = len(client_names)
+
+    for r in range(1, args.warmup + 1):
+        one_round(r)
+
+    ctx.barrier()
+    if device_is_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for r in range(args.warmup + 1, args.warmup + args.steps + 1):
+        one_round(r)
+    ctx.barrier()
+    if device_is_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    elapsed = ctx.all_reduce_scalar(elapsed, op="max")
+
+    if ctx.is_rank0():
+        total_images = imgs_per_client * n_clients * args.steps
+        value = total_images / elapsed
+        result = {
+            "metric": "images/sec (whole node) per fed round, FedSTIL ResNet50",
+            "value": round(value, 2),
+            "unit": "images/sec",
+            "n_gpus": max(1, ctx.world_size),
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if device_is_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"{args.model}-{args.method}",
+                "global_batch": args.batch * n_clients,
+                "img": args.img,
+                "images_per_round_per_client": imgs_per_client,
+                "num_classes": args.num_classes,
+                "parallelism": f"fed-dp{max(1, ctx.world_size)} (1 client/GPU)",
+            },
+        }
+        print(json.dumps(result))
+    destroy_context()
+
+
+if __name__ == "__main__":
+    main()

@@ -72,14 +72,10 @@ class ImportanceModel(ModelModule):
                     loss = self.operator._invoke_train(self, data, target)["loss"]
                 loss.backward()
                 scale = len(data) / n_batches
-                grads = {n: (p.grad.detach().float() * scale if self.mode == "sq"
-                             else p.grad.detach().abs() * scale)
+                grads = {n: p.grad.detach()
                          for n, p in self.params.items() if p.grad is not None}
-                if self.mode == "sq":
-                    ops.importance_update(precision, grads, mode="sq")
-                else:  # abs grads pre-scaled; plain add
-                    for n, g in grads.items():
-                        precision[n] += g
+                ops.importance_update(precision, grads, mode=self.mode,
+                                      scale=scale)
         self.net.zero_grad(set_to_none=True)
         return precision
 

@@ -40,7 +40,18 @@ class ClientModule:
         self.operator.logger = self.logger
 
     # ------------------------------------------------------------------ ckpt
+    # FLREID_DISABLE_CKPT=1 turns the per-round ckpt audit trail off for
+    # benchmarking (the reference's disk round trip per dispatch/upload is an
+    # auditing feature, not training semantics — ref:experiment.py:199-202)
+    @staticmethod
+    def _ckpt_disabled() -> bool:
+        return os.environ.get("FLREID_DISABLE_CKPT", "0") == "1"
+
     def load_state(self, state_name: str, default_value: Any = None) -> Any:
+        if self._ckpt_disabled():
+            if default_value is not None:
+                return default_value
+            raise ValueError("ckpt disabled and no default value")
         state_path = os.path.join(self.ckpt_path, f"{state_name}.ckpt")
         os.makedirs(self.ckpt_path, exist_ok=True)
         if os.path.exists(state_path):
@@ -50,7 +61,7 @@ class ClientModule:
         raise ValueError(f"State checkpoint does not exist in '{state_path}'.")
 
     def save_state(self, state_name: str, state: Any, cover: bool = False) -> None:
-        if state_name is None:
+        if state_name is None or self._ckpt_disabled():
             return
         state_path = os.path.join(self.ckpt_path, f"{state_name}.ckpt")
         os.makedirs(self.ckpt_path, exist_ok=True)

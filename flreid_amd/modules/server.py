@@ -41,6 +41,10 @@ class ServerModule:
 
     # ------------------------------------------------------------------ ckpt
     def load_state(self, state_name: str, default_value: Any = None) -> Any:
+        if os.environ.get("FLREID_DISABLE_CKPT", "0") == "1":
+            if default_value is not None:
+                return default_value
+            raise ValueError("ckpt disabled and no default value")
         state_path = os.path.join(self.ckpt_path, f"{state_name}.ckpt")
         os.makedirs(self.ckpt_path, exist_ok=True)
         if os.path.exists(state_path):
@@ -50,7 +54,7 @@ class ServerModule:
         raise ValueError(f"State checkpoint does not exist in '{state_path}'.")
 
     def save_state(self, state_name: str, state: Any, cover: bool = False) -> None:
-        if not _is_writer_rank():
+        if not _is_writer_rank() or os.environ.get("FLREID_DISABLE_CKPT", "0") == "1":
             return
         state_path = os.path.join(self.ckpt_path, f"{state_name}.ckpt")
         os.makedirs(self.ckpt_path, exist_ok=True)

@@ -2,19 +2,19 @@
 
 CPU tensors  -> pure-PyTorch reference implementations (ops/reference.py).
 CUDA tensors -> hand-written HIP/CDNA4 kernels from the in-tree extension
-                (ops/csrc, built to flreid_amd/ops/_flreid_hip.so by
-                `python setup.py build_ext --inplace` or __graft_entry__.build()).
+                flreid_amd/ops/_flreid_hip.so (sources in ops/csrc, built by
+                `python -m flreid_amd.ops.build` / __graft_entry__.build()).
 
-On a GPU box the HIP path is mandatory: if the extension is missing we raise
-instead of silently falling back to eager PyTorch (set FLREID_ALLOW_EAGER=1 to
-override for debugging).  This keeps "gpu tests green" honest — they can only
-pass through the native kernels.
+On a GPU box the HIP path is mandatory: if the extension is missing the op
+raises instead of silently falling back to eager PyTorch (set
+FLREID_ALLOW_EAGER=1 to debug).  Numerics ground truth for every kernel is
+the fp32 reference implementation (tests/test_ops_gpu.py).
 """
 
 from __future__ import annotations
 
 import os
-from typing import Optional
+from typing import Dict, Optional
 
 import torch
 
@@ -23,16 +23,18 @@ from flreid_amd.ops import reference as ref
 _EXT = None
 _EXT_ERR: Optional[str] = None
 
+_F32, _BF16 = 0, 1
+
 
 def _load_extension():
     global _EXT, _EXT_ERR
     if _EXT is not None or _EXT_ERR is not None:
         return _EXT
     try:
-        import importlib
+        from flreid_amd.ops import _flreid_hip  # built in-tree
 
-        _EXT = importlib.import_module("flreid_amd.ops._flreid_hip")
-    except Exception as e:  # pragma: no cover - exercised only on GPU boxes
+        _EXT = _flreid_hip
+    except Exception as e:  # pragma: no cover - GPU boxes only
         _EXT_ERR = f"{type(e).__name__}: {e}"
         _EXT = None
     return _EXT
@@ -42,105 +44,220 @@ def extension_available() -> bool:
     return _load_extension() is not None
 
 
-def _gpu_impl(opname: str):
-    """Return the extension entry point for `opname` or raise loudly."""
+def _ext_or_raise(opname: str):
     ext = _load_extension()
-    if ext is not None and hasattr(ext, opname):
-        return getattr(ext, opname)
+    if ext is not None:
+        return ext
     if os.environ.get("FLREID_ALLOW_EAGER", "0") == "1":
         return None
     raise RuntimeError(
-        f"flreid HIP extension does not provide '{opname}' on this GPU "
-        f"(extension={'loaded' if ext is not None else f'missing: {_EXT_ERR}'}). "
-        "Build it with `python setup.py build_ext --inplace` "
-        "(PYTORCH_ROCM_ARCH=gfx950) or set FLREID_ALLOW_EAGER=1 to debug with "
-        "eager PyTorch."
+        f"flreid HIP extension missing for GPU op '{opname}' ({_EXT_ERR}). "
+        "Build it with `python -m flreid_amd.ops.build` "
+        "or set FLREID_ALLOW_EAGER=1 to debug with eager PyTorch."
     )
 
 
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _dt(t: torch.Tensor) -> int:
+    if t.dtype == torch.float32:
+        return _F32
+    if t.dtype == torch.bfloat16:
+        return _BF16
+    raise TypeError(f"unsupported dtype {t.dtype}")
+
+
 # ---------------------------------------------------------------------------
-# public ops — each dispatches on device
+# distances / similarity (K8)
 # ---------------------------------------------------------------------------
 
+def _pairwise_gpu(ext, a: torch.Tensor, b: torch.Tensor, mode: int) -> torch.Tensor:
+    a = a.contiguous().float()
+    b = b.contiguous().float()
+    m, n, d = a.shape[0], b.shape[0], a.shape[1]
+    out = torch.empty(m, n, device=a.device, dtype=torch.float32)
+    if mode == 2:
+        aa = torch.empty(m, device=a.device, dtype=torch.float32)
+        bb = torch.empty(n, device=a.device, dtype=torch.float32)
+        ext.rowsq(a.data_ptr(), aa.data_ptr(), m, d, _stream())
+        ext.rowsq(b.data_ptr(), bb.data_ptr(), n, d, _stream())
+    else:
+        aa = bb = out  # unused
+    ext.pairwise(a.data_ptr(), b.data_ptr(), aa.data_ptr(), bb.data_ptr(),
+                 out.data_ptr(), m, n, d, mode, _stream())
+    return out
+
+
+def similarity_matrix(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """A·Bᵀ — the eval GEMM."""
+    if a.is_cuda and not a.requires_grad and not b.requires_grad:
+        ext = _ext_or_raise("pairwise")
+        if ext is not None:
+            return _pairwise_gpu(ext, a, b, 0)
+    return a.float() @ b.float().t()
+
+
 def pairwise_sqeuclidean(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    if a.is_cuda:
-        fn = _gpu_impl("pairwise_sqeuclidean")
-        if fn is not None:
-            return fn(a.contiguous(), b.contiguous())
+    if a.is_cuda and not a.requires_grad and not b.requires_grad:
+        ext = _ext_or_raise("pairwise")
+        if ext is not None:
+            return _pairwise_gpu(ext, a, b, 2)
     return ref.pairwise_sqeuclidean(a, b)
 
 
 def pairwise_cosine_distance(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    if a.is_cuda:
-        fn = _gpu_impl("pairwise_cosine_distance")
-        if fn is not None:
-            return fn(a.contiguous(), b.contiguous())
+    if a.is_cuda and not a.requires_grad and not b.requires_grad:
+        ext = _ext_or_raise("pairwise")
+        if ext is not None:
+            return _pairwise_gpu(ext, l2_normalize(a.float()),
+                                 l2_normalize(b.float()), 1)
     return ref.pairwise_cosine_distance(a, b)
 
 
+# ---------------------------------------------------------------------------
+# rowwise L2 normalize (K11)
+# ---------------------------------------------------------------------------
+
 def l2_normalize(x: torch.Tensor, dim: int = 1) -> torch.Tensor:
-    if x.is_cuda and dim in (1, -1) and x.dim() == 2 and not x.requires_grad:
-        fn = _gpu_impl("l2_normalize")
-        if fn is not None:
-            return fn(x.contiguous())
+    if (x.is_cuda and x.dim() == 2 and dim in (1, -1)
+            and not x.requires_grad and x.dtype in (torch.float32, torch.bfloat16)):
+        ext = _ext_or_raise("l2norm_rows")
+        if ext is not None:
+            x = x.contiguous()
+            y = torch.empty_like(x)
+            ext.l2norm_rows(x.data_ptr(), y.data_ptr(), x.shape[0], x.shape[1],
+                            _dt(x), 1e-12, _stream())
+            return y
     return ref.l2_normalize(x, dim=dim)
 
 
-def kl_distance(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    # tiny tensors (task tokens) — the reference impl is fine on any device
-    return ref.kl_distance(a, b)
-
+# ---------------------------------------------------------------------------
+# fused label-smooth CE (K6)
+# ---------------------------------------------------------------------------
 
 class _CeLabelSmoothFn(torch.autograd.Function):
-    """Fused label-smooth CE on GPU (fwd computes loss + saves softmax-grad)."""
-
     @staticmethod
-    def forward(ctx, score, target, epsilon):
-        fn = _gpu_impl("ce_label_smooth_fwd")
-        if fn is None:
-            raise RuntimeError("unreachable: eager fallback handled in wrapper")
-        loss, grad = fn(score.contiguous(), target.contiguous(), float(epsilon))
+    def forward(ctx, score, target, epsilon, ext):
+        score_c = score.contiguous()
+        B, C = score_c.shape
+        row_loss = torch.empty(B, device=score.device, dtype=torch.float32)
+        grad = torch.empty_like(score_c)
+        ext.ce_smooth(score_c.data_ptr(), target.contiguous().data_ptr(),
+                      row_loss.data_ptr(), grad.data_ptr(), B, C,
+                      _dt(score_c), float(epsilon), _stream())
         ctx.save_for_backward(grad)
-        return loss
+        return row_loss.sum() * (1.0 / B)
 
     @staticmethod
     def backward(ctx, grad_out):
         (grad,) = ctx.saved_tensors
-        return grad * grad_out, None, None
+        return grad * grad_out, None, None, None
 
 
 def ce_label_smooth(score: torch.Tensor, target: torch.Tensor,
                     epsilon: float = 0.1) -> torch.Tensor:
-    if score.is_cuda:
-        try:
-            _gpu_impl("ce_label_smooth_fwd")
-            return _CeLabelSmoothFn.apply(score, target, epsilon)
-        except RuntimeError:
-            if os.environ.get("FLREID_ALLOW_EAGER", "0") != "1":
-                raise
+    if score.is_cuda and score.dim() == 2:
+        ext = _ext_or_raise("ce_smooth")
+        if ext is not None:
+            return _CeLabelSmoothFn.apply(score, target.long(), epsilon, ext)
     return ref.ce_label_smooth(score, target, epsilon)
+
+
+# ---------------------------------------------------------------------------
+# FedSTIL composition (K1/K2 prologue, standalone form)
+# ---------------------------------------------------------------------------
+
+class _ComposeFn(torch.autograd.Function):
+    """theta = atten ⊙_lastdim gw + aw.  gw/atten frozen in FedSTIL; grad
+    flows to aw (identity) and, when atten trains (fedstil-atten), reduces
+    over all dims but the last."""
+
+    @staticmethod
+    def forward(ctx, gw, atten, aw, ext):
+        gw_c, aw_c = gw.contiguous(), aw.contiguous()
+        at = atten.detach().contiguous().float()
+        out = torch.empty_like(gw_c)
+        ext.compose(gw_c.data_ptr(), at.data_ptr(), aw_c.data_ptr(),
+                    out.data_ptr(), gw_c.numel(), gw_c.shape[-1], _dt(gw_c),
+                    _stream())
+        ctx.atten_requires = atten.requires_grad
+        if ctx.atten_requires:
+            ctx.save_for_backward(gw_c)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        g_atten = None
+        if ctx.atten_requires:
+            (gw,) = ctx.saved_tensors
+            g_atten = (grad_out * gw).reshape(-1, gw.shape[-1]).sum(dim=0)
+        return None, g_atten, grad_out, None
+
+
+def adaptive_compose(global_weight: torch.Tensor, atten: torch.Tensor,
+                     adaptive_weight: torch.Tensor) -> torch.Tensor:
+    if (global_weight.is_cuda and not global_weight.requires_grad
+            and global_weight.dtype in (torch.float32, torch.bfloat16)
+            and adaptive_weight.dtype == global_weight.dtype):
+        ext = _ext_or_raise("compose")
+        if ext is not None:
+            return _ComposeFn.apply(global_weight, atten, adaptive_weight, ext)
+    return ref.adaptive_compose(global_weight, atten, adaptive_weight)
+
+
+# ---------------------------------------------------------------------------
+# EWC/MAS importance accumulation (K9)
+# ---------------------------------------------------------------------------
+
+def importance_update(importance: Dict[str, torch.Tensor],
+                      grads: Dict[str, torch.Tensor], mode: str = "sq",
+                      scale: float = 1.0) -> None:
+    first = next(iter(importance.values()), None)
+    if first is not None and first.is_cuda:
+        ext = _ext_or_raise("importance")
+        if ext is not None:
+            for n, g in grads.items():
+                if g is None:
+                    continue
+                F = importance[n]
+                ext.importance(F.data_ptr(), g.contiguous().data_ptr(),
+                               g.numel(), _dt(g), mode == "sq", scale,
+                               _stream())
+            return
+    for n, g in grads.items():
+        if g is None:
+            continue
+        if mode == "sq":
+            importance[n] += (g.detach().float() ** 2) * scale
+        else:
+            importance[n] += g.detach().float().abs() * scale
+
+
+# ---------------------------------------------------------------------------
+# pass-throughs (small tensors / composed autograd paths)
+# ---------------------------------------------------------------------------
+
+kl_distance = ref.kl_distance
+kd_loss = ref.kd_loss
+quadratic_penalty = ref.quadratic_penalty
+l1_drift = ref.l1_drift
 
 
 def triplet_loss(feature: torch.Tensor, target: torch.Tensor,
                  margin: Optional[float] = 0.3, norm_feat: bool = False,
                  hard_mining: bool = True) -> torch.Tensor:
-    # The fused GPU kernel covers the inference-free hot case; autograd path
-    # composes distance + mining ops that are themselves dispatched.
     return ref.triplet_loss(feature, target, margin, norm_feat, hard_mining)
-
-
-def kd_loss(logits_student: torch.Tensor, logits_teacher: torch.Tensor,
-            temperature: float = 4.0) -> torch.Tensor:
-    return ref.kd_loss(logits_student, logits_teacher, temperature)
-
-
-adaptive_compose = ref.adaptive_compose
-importance_update = ref.importance_update
-quadratic_penalty = ref.quadratic_penalty
-l1_drift = ref.l1_drift
 
 
 def cmc_map(query_features, query_labels, gallery_features, gallery_labels,
             query_camera_labels=None, gallery_camera_labels=None):
+    """On GPU the Q×G similarity runs on the MFMA pairwise kernel; the rank
+    statistics stay in (device-side) torch ops."""
+    if query_features.is_cuda:
+        sims = similarity_matrix(query_features, gallery_features)
+        return ref.cmc_map_from_sims(sims, query_labels, gallery_labels,
+                                     query_camera_labels, gallery_camera_labels)
     return ref.cmc_map(query_features, query_labels, gallery_features,
                        gallery_labels, query_camera_labels, gallery_camera_labels)

@@ -1,0 +1,57 @@
+"""In-tree build of the MI355X HIP extension.
+
+`python -m flreid_amd.ops.build` (or __graft_entry__.build()) compiles
+ops/csrc/*.{hip,cpp} with hipcc for gfx950 into flreid_amd/ops/_flreid_hip.so.
+The built .so travels to GPU boxes with the repo snapshot.
+"""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import sysconfig
+
+CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)), "csrc")
+OUT = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_flreid_hip.so")
+
+SOURCES = ["module.cpp", "elementwise.hip", "distance.hip"]
+
+
+def _pybind11_includes():
+    import pybind11
+    return [pybind11.get_include()]
+
+
+def needs_rebuild() -> bool:
+    if not os.path.exists(OUT):
+        return True
+    out_mtime = os.path.getmtime(OUT)
+    for f in os.listdir(CSRC):
+        if os.path.getmtime(os.path.join(CSRC, f)) > out_mtime:
+            return True
+    return False
+
+
+def build(force: bool = False, arch: str = "gfx950", verbose: bool = True) -> str:
+    if not force and not needs_rebuild():
+        if verbose:
+            print(f"[flreid build] up to date: {OUT}")
+        return OUT
+    hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+    includes = [sysconfig.get_paths()["include"]] + _pybind11_includes()
+    cmd = [
+        hipcc, f"--offload-arch={arch}", "-O3", "-std=c++17", "-fPIC",
+        "-shared", "-fvisibility=hidden",
+        *[f"-I{p}" for p in includes],
+        *[os.path.join(CSRC, s) for s in SOURCES],
+        "-o", OUT,
+    ]
+    if verbose:
+        print("[flreid build]", " ".join(cmd))
+    subprocess.run(cmd, check=True)
+    return OUT
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)

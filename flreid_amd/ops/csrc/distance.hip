@@ -1,0 +1,137 @@
+// Pairwise distance / similarity GEMM on MFMA (K8 in SURVEY.md §2.9).
+//
+// Computes OUT[m, n] over row-major feature matrices A[M, D], B[N, D]:
+//   mode 0:  A·Bᵀ                      (similarity, the CMC/mAP eval GEMM —
+//                                       replaces the per-query GEMV loop of
+//                                       ref:tools/evaluate.py:103-142)
+//   mode 1:  1 − A·Bᵀ                  (cosine distance on pre-normalised
+//                                       rows, ref:tools/distance.py:19-30)
+//   mode 2:  aa[m] + bb[n] − 2·A·Bᵀ    (squared euclidean,
+//                                       ref:tools/distance.py:9-16)
+//
+// Exact fp32 numerics via the f32-input MFMA (v_mfma_f32_32x32x2_f32): the
+// result is bit-for-bit an fmaf chain (guide §3), so eval metrics match the
+// fp32 reference exactly up to summation order.
+//
+// Structure (guide §5 canonical): 64×64 block tile, 4 waves × one 32×32
+// sub-tile, LDS-staged K panels (BK=32, padded rows: conflict-free b32 reads).
+
+#include "common.h"
+
+namespace flreid {
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+constexpr int BM = 64;      // block tile rows
+constexpr int BN = 64;      // block tile cols
+constexpr int BK = 32;      // K panel
+constexpr int PAD = 1;      // LDS row padding (33 floats/row)
+
+template <int MODE>
+__global__ __launch_bounds__(256) void pairwise_mfma_kernel(
+    const float* __restrict__ A, const float* __restrict__ B,
+    const float* __restrict__ aa, const float* __restrict__ bb,
+    float* __restrict__ OUT, int64_t M, int64_t N, int64_t D) {
+  __shared__ float ldsA[BM][BK + PAD];
+  __shared__ float ldsB[BN][BK + PAD];
+
+  const int64_t row0 = (int64_t)blockIdx.x * BM;
+  const int64_t col0 = (int64_t)blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;            // 4 waves
+  const int wr = (wave >> 1) * 32;      // wave sub-tile row offset (0/32)
+  const int wc = (wave & 1) * 32;       // wave sub-tile col offset (0/32)
+
+  f32x16 acc = {};
+
+  const int lc = tid & (BK - 1);        // load col 0..31
+  const int lr0 = tid >> 5;             // load row base 0..7 (8 rows/pass)
+
+  for (int64_t k0 = 0; k0 < D; k0 += BK) {
+#pragma unroll
+    for (int r = 0; r < BM; r += 8) {
+      const int lr = lr0 + r;
+      const int64_t ar = row0 + lr;
+      const int64_t bc = col0 + lr;
+      const int64_t kk = k0 + lc;
+      ldsA[lr][lc] = (ar < M && kk < D) ? A[ar * D + kk] : 0.0f;
+      ldsB[lr][lc] = (bc < N && kk < D) ? B[bc * D + kk] : 0.0f;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const float a = ldsA[wr + (lane & 31)][kk + (lane >> 5)];
+      const float b = ldsB[wc + (lane & 31)][kk + (lane >> 5)];
+      acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // C/D mapping (guide §3): col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  const int jc = col0 + wc + (lane & 31);
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int ir = row0 + wr + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+    if (ir < M && jc < N) {
+      float v = acc[reg];
+      if (MODE == 1) v = 1.0f - v;
+      if (MODE == 2) v = aa[ir] + bb[jc] - 2.0f * v;
+      OUT[(int64_t)ir * N + jc] = v;
+    }
+  }
+}
+
+extern "C" void flreid_pairwise(const float* A, const float* B,
+                                const float* aa, const float* bb, float* OUT,
+                                int64_t M, int64_t N, int64_t D, int mode,
+                                hipStream_t stream) {
+  dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)((N + BN - 1) / BN));
+  dim3 block(256);
+  switch (mode) {
+    case 0:
+      hipLaunchKernelGGL((pairwise_mfma_kernel<0>), grid, block, 0, stream, A,
+                         B, aa, bb, OUT, M, N, D);
+      break;
+    case 1:
+      hipLaunchKernelGGL((pairwise_mfma_kernel<1>), grid, block, 0, stream, A,
+                         B, aa, bb, OUT, M, N, D);
+      break;
+    case 2:
+      hipLaunchKernelGGL((pairwise_mfma_kernel<2>), grid, block, 0, stream, A,
+                         B, aa, bb, OUT, M, N, D);
+      break;
+    default:
+      throw std::runtime_error("bad pairwise mode");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// row sums of squares (for the euclidean epilogue)
+template <int BLOCK>
+__global__ void rowsq_kernel(const float* __restrict__ x,
+                             float* __restrict__ out, int64_t rows,
+                             int64_t cols) {
+  __shared__ float scratch[BLOCK / kWave];
+  const int64_t row = blockIdx.x;
+  if (row >= rows) return;
+  const float* xr = x + row * cols;
+  float ss = 0.f;
+  for (int64_t c = threadIdx.x; c < cols; c += BLOCK) {
+    const float v = xr[c];
+    ss += v * v;
+  }
+  ss = block_reduce_sum<BLOCK>(ss, scratch);
+  if (threadIdx.x == 0) out[row] = ss;
+}
+
+extern "C" void flreid_rowsq(const float* x, float* out, int64_t rows,
+                             int64_t cols, hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  hipLaunchKernelGGL((rowsq_kernel<BLOCK>), dim3((unsigned)rows), dim3(BLOCK),
+                     0, stream, x, out, rows, cols);
+  HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace flreid

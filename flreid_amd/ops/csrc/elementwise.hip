@@ -1,0 +1,206 @@
+// Fused elementwise / row-reduction kernels for the ReID hot path.
+//
+//  - l2norm_rows:  y = x / max(||x||, eps) rowwise (K11 in SURVEY.md §2.9;
+//                  replaces F.normalize in the eval feature path,
+//                  ref:methods/fedavg.py:158-168)
+//  - ce_label_smooth: fused log-softmax + smoothed NLL + analytic grad in one
+//                  pass over the [B, C] score matrix (K6; the reference
+//                  round-tripped a one-hot through the host every batch,
+//                  ref:criterions/cross_entropy.py:36-38)
+//  - adaptive_compose: theta = atten (bcast over last dim) * gw + aw (the
+//                  FedSTIL composition, K1-prologue candidate; standalone
+//                  fused form saves two full tensor passes vs mul+add)
+//  - importance_sq/abs: F += g^2 or |g| (K9, EWC/MAS accumulation)
+//
+// All kernels: fp32 compute; bf16 or fp32 I/O.
+
+#include "common.h"
+
+namespace flreid {
+
+// --------------------------------------------------------------------------
+// rowwise L2 normalize
+// --------------------------------------------------------------------------
+
+template <typename T, int BLOCK>
+__global__ void l2norm_rows_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   int64_t rows, int64_t cols, float eps) {
+  __shared__ float scratch[BLOCK / kWave];
+  const int64_t row = blockIdx.x;
+  if (row >= rows) return;
+  const T* xr = x + row * cols;
+  T* yr = y + row * cols;
+
+  float ss = 0.f;
+  for (int64_t c = threadIdx.x; c < cols; c += BLOCK) {
+    const float v = load_as_float(xr, c);
+    ss += v * v;
+  }
+  const float total = block_reduce_sum<BLOCK>(ss, scratch);
+  const float inv = 1.0f / fmaxf(sqrtf(total), eps);
+  for (int64_t c = threadIdx.x; c < cols; c += BLOCK) {
+    store_from_float(yr, c, load_as_float(xr, c) * inv);
+  }
+}
+
+extern "C" void flreid_l2norm_rows(const void* x, void* y, int64_t rows,
+                                   int64_t cols, int dtype, float eps,
+                                   hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  dim3 grid((unsigned)rows), block(BLOCK);
+  if (dtype == kF32) {
+    hipLaunchKernelGGL((l2norm_rows_kernel<float, BLOCK>), grid, block, 0,
+                       stream, (const float*)x, (float*)y, rows, cols, eps);
+  } else {
+    hipLaunchKernelGGL((l2norm_rows_kernel<__hip_bfloat16, BLOCK>), grid,
+                       block, 0, stream, (const __hip_bfloat16*)x,
+                       (__hip_bfloat16*)y, rows, cols, eps);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// --------------------------------------------------------------------------
+// fused label-smooth CE: per-row loss + grad
+//   loss_b = sum_c -t_bc * logp_bc,  t = (1-eps)*onehot + eps/C
+//   grad_bc = (softmax_bc - t_bc) / B          (d(mean_b sum_c)/dscore)
+// --------------------------------------------------------------------------
+
+template <typename T, int BLOCK>
+__global__ void ce_smooth_kernel(const T* __restrict__ score,
+                                 const int64_t* __restrict__ target,
+                                 float* __restrict__ row_loss,
+                                 T* __restrict__ grad, int64_t B, int64_t C,
+                                 float eps_smooth, float inv_B) {
+  __shared__ float scratch[BLOCK / kWave];
+  const int64_t b = blockIdx.x;
+  if (b >= B) return;
+  const T* s = score + b * C;
+  T* g = grad + b * C;
+  const int64_t y = target[b];
+  const float uni = eps_smooth / (float)C;
+  const float on = 1.0f - eps_smooth;
+
+  // pass 1: max
+  float m = -INFINITY;
+  for (int64_t c = threadIdx.x; c < C; c += BLOCK) {
+    m = fmaxf(m, load_as_float(s, c));
+  }
+  m = block_reduce_max(m, scratch, BLOCK);
+
+  // pass 2: sum exp
+  float se = 0.f;
+  for (int64_t c = threadIdx.x; c < C; c += BLOCK) {
+    se += __expf(load_as_float(s, c) - m);
+  }
+  se = block_reduce_sum<BLOCK>(se, scratch);
+  const float log_z = __logf(se) + m;
+  const float inv_se = 1.0f / se;
+
+  // pass 3: loss + grad
+  float loss = 0.f;
+  for (int64_t c = threadIdx.x; c < C; c += BLOCK) {
+    const float sv = load_as_float(s, c);
+    const float logp = sv - log_z;
+    const float p = __expf(sv - m) * inv_se;
+    const float t = uni + (c == y ? on : 0.0f);
+    loss += -t * logp;
+    store_from_float(g, c, (p - t) * inv_B);
+  }
+  loss = block_reduce_sum<BLOCK>(loss, scratch);
+  if (threadIdx.x == 0) row_loss[b] = loss;
+}
+
+extern "C" void flreid_ce_smooth(const void* score, const int64_t* target,
+                                 float* row_loss, void* grad, int64_t B,
+                                 int64_t C, int dtype, float eps_smooth,
+                                 hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  dim3 grid((unsigned)B), block(BLOCK);
+  const float inv_B = 1.0f / (float)B;
+  if (dtype == kF32) {
+    hipLaunchKernelGGL((ce_smooth_kernel<float, BLOCK>), grid, block, 0,
+                       stream, (const float*)score, target, row_loss,
+                       (float*)grad, B, C, eps_smooth, inv_B);
+  } else {
+    hipLaunchKernelGGL((ce_smooth_kernel<__hip_bfloat16, BLOCK>), grid, block,
+                       0, stream, (const __hip_bfloat16*)score, target,
+                       row_loss, (__hip_bfloat16*)grad, B, C, eps_smooth,
+                       inv_B);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// --------------------------------------------------------------------------
+// FedSTIL composition: theta = atten[i % L] * gw + aw (atten over last dim)
+// --------------------------------------------------------------------------
+
+template <typename T>
+__global__ void compose_kernel(const T* __restrict__ gw,
+                               const float* __restrict__ atten,
+                               const T* __restrict__ aw, T* __restrict__ out,
+                               int64_t numel, int64_t L) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= numel) return;
+  const float a = atten[i % L];
+  store_from_float(out, i,
+                   fmaf(a, load_as_float(gw, i), load_as_float(aw, i)));
+}
+
+extern "C" void flreid_compose(const void* gw, const float* atten,
+                               const void* aw, void* out, int64_t numel,
+                               int64_t L, int dtype, hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  dim3 grid((unsigned)((numel + BLOCK - 1) / BLOCK)), block(BLOCK);
+  if (dtype == kF32) {
+    hipLaunchKernelGGL((compose_kernel<float>), grid, block, 0, stream,
+                       (const float*)gw, atten, (const float*)aw, (float*)out,
+                       numel, L);
+  } else {
+    hipLaunchKernelGGL((compose_kernel<__hip_bfloat16>), grid, block, 0,
+                       stream, (const __hip_bfloat16*)gw, atten,
+                       (const __hip_bfloat16*)aw, (__hip_bfloat16*)out, numel,
+                       L);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// --------------------------------------------------------------------------
+// EWC/MAS importance accumulation: F += g*g (sq) or F += |g| * scale
+// --------------------------------------------------------------------------
+
+template <typename T, bool SQ>
+__global__ void importance_kernel(float* __restrict__ F,
+                                  const T* __restrict__ g, int64_t numel,
+                                  float scale) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= numel) return;
+  const float gv = load_as_float(g, i);
+  F[i] += SQ ? gv * gv * scale : fabsf(gv) * scale;
+}
+
+extern "C" void flreid_importance(float* F, const void* g, int64_t numel,
+                                  int dtype, int sq, float scale,
+                                  hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  dim3 grid((unsigned)((numel + BLOCK - 1) / BLOCK)), block(BLOCK);
+  if (dtype == kF32) {
+    if (sq)
+      hipLaunchKernelGGL((importance_kernel<float, true>), grid, block, 0,
+                         stream, F, (const float*)g, numel, scale);
+    else
+      hipLaunchKernelGGL((importance_kernel<float, false>), grid, block, 0,
+                         stream, F, (const float*)g, numel, scale);
+  } else {
+    if (sq)
+      hipLaunchKernelGGL((importance_kernel<__hip_bfloat16, true>), grid,
+                         block, 0, stream, F, (const __hip_bfloat16*)g, numel,
+                         scale);
+    else
+      hipLaunchKernelGGL((importance_kernel<__hip_bfloat16, false>), grid,
+                         block, 0, stream, F, (const __hip_bfloat16*)g, numel,
+                         scale);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace flreid

@@ -1,0 +1,84 @@
+// Python bindings for the flreid MI355X kernels.
+//
+// Deliberately torch-header-free: tensors cross as raw device pointers plus
+// the caller's torch HIP stream (uintptr_t), so the extension compiles with
+// plain hipcc for gfx950 — no hipify, no CUDA-compat layer — and launches
+// stay ordered with PyTorch work on the same stream.
+
+#include <pybind11/pybind11.h>
+
+#include <cstdint>
+
+#include <hip/hip_runtime.h>
+
+namespace flreid {
+extern "C" void flreid_l2norm_rows(const void*, void*, int64_t, int64_t, int,
+                                   float, hipStream_t);
+extern "C" void flreid_ce_smooth(const void*, const int64_t*, float*, void*,
+                                 int64_t, int64_t, int, float, hipStream_t);
+extern "C" void flreid_compose(const void*, const float*, const void*, void*,
+                               int64_t, int64_t, int, hipStream_t);
+extern "C" void flreid_importance(float*, const void*, int64_t, int, int,
+                                  float, hipStream_t);
+extern "C" void flreid_pairwise(const float*, const float*, const float*,
+                                const float*, float*, int64_t, int64_t,
+                                int64_t, int, hipStream_t);
+extern "C" void flreid_rowsq(const float*, float*, int64_t, int64_t,
+                             hipStream_t);
+}  // namespace flreid
+
+namespace py = pybind11;
+
+static hipStream_t as_stream(uintptr_t s) {
+  return reinterpret_cast<hipStream_t>(s);
+}
+
+PYBIND11_MODULE(_flreid_hip, m) {
+  m.doc() = "flreid MI355X (gfx950) kernels";
+
+  m.def("l2norm_rows",
+        [](uintptr_t x, uintptr_t y, int64_t rows, int64_t cols, int dtype,
+           float eps, uintptr_t stream) {
+          flreid::flreid_l2norm_rows((const void*)x, (void*)y, rows, cols,
+                                     dtype, eps, as_stream(stream));
+        });
+
+  m.def("ce_smooth",
+        [](uintptr_t score, uintptr_t target, uintptr_t row_loss,
+           uintptr_t grad, int64_t B, int64_t C, int dtype, float eps,
+           uintptr_t stream) {
+          flreid::flreid_ce_smooth((const void*)score, (const int64_t*)target,
+                                   (float*)row_loss, (void*)grad, B, C, dtype,
+                                   eps, as_stream(stream));
+        });
+
+  m.def("compose",
+        [](uintptr_t gw, uintptr_t atten, uintptr_t aw, uintptr_t out,
+           int64_t numel, int64_t L, int dtype, uintptr_t stream) {
+          flreid::flreid_compose((const void*)gw, (const float*)atten,
+                                 (const void*)aw, (void*)out, numel, L, dtype,
+                                 as_stream(stream));
+        });
+
+  m.def("importance",
+        [](uintptr_t F, uintptr_t g, int64_t numel, int dtype, bool sq,
+           float scale, uintptr_t stream) {
+          flreid::flreid_importance((float*)F, (const void*)g, numel, dtype,
+                                    sq ? 1 : 0, scale, as_stream(stream));
+        });
+
+  m.def("pairwise",
+        [](uintptr_t A, uintptr_t B, uintptr_t aa, uintptr_t bb, uintptr_t out,
+           int64_t M, int64_t N, int64_t D, int mode, uintptr_t stream) {
+          flreid::flreid_pairwise((const float*)A, (const float*)B,
+                                  (const float*)aa, (const float*)bb,
+                                  (float*)out, M, N, D, mode,
+                                  as_stream(stream));
+        });
+
+  m.def("rowsq", [](uintptr_t x, uintptr_t out, int64_t rows, int64_t cols,
+                    uintptr_t stream) {
+    flreid::flreid_rowsq((const float*)x, (float*)out, rows, cols,
+                         as_stream(stream));
+  });
+}

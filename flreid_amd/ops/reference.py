@@ -119,19 +119,19 @@ def kd_loss(logits_student: torch.Tensor, logits_teacher: torch.Tensor,
 
 def importance_update(importance: Dict[str, torch.Tensor],
                       grads: Dict[str, torch.Tensor],
-                      mode: str = "sq") -> None:
+                      mode: str = "sq", scale: float = 1.0) -> None:
     """In-place accumulate per-parameter importance.
 
-    mode='sq'  : F += g²   (EWC Fisher, ref:methods/ewc.py:56-78)
-    mode='abs' : F += |g|  (MAS importance, ref:methods/mas.py:72-74)
+    mode='sq'  : F += g²·scale   (EWC Fisher, ref:methods/ewc.py:56-78)
+    mode='abs' : F += |g|·scale  (MAS importance, ref:methods/mas.py:72-74)
     """
     for n, g in grads.items():
         if g is None:
             continue
         if mode == "sq":
-            importance[n] += g.detach() ** 2
+            importance[n] += (g.detach().float() ** 2) * scale
         elif mode == "abs":
-            importance[n] += g.detach().abs()
+            importance[n] += g.detach().float().abs() * scale
         else:
             raise ValueError(mode)
 
@@ -202,12 +202,21 @@ def cmc_map(query_features: torch.Tensor, query_labels: torch.Tensor,
     Ties in similarity are broken by descending gallery index (matching
     np.argsort(sim)[::-1] stable-sort reversal).
     """
-    q, g = query_features.size(0), gallery_features.size(0)
-    device = query_features.device
+    sims = query_features.float() @ gallery_features.float().t()     # [Q, G]
+    return cmc_map_from_sims(sims, query_labels, gallery_labels,
+                             query_camera_labels, gallery_camera_labels)
+
+
+@torch.no_grad()
+def cmc_map_from_sims(sims: torch.Tensor, query_labels: torch.Tensor,
+                      gallery_labels: torch.Tensor,
+                      query_camera_labels: Optional[torch.Tensor] = None,
+                      gallery_camera_labels: Optional[torch.Tensor] = None,
+                      ) -> Tuple[torch.Tensor, float]:
+    q, g = sims.shape
+    device = sims.device
     ql = query_labels.to(device).view(q, 1)
     gl = gallery_labels.to(device).view(1, g)
-
-    sims = query_features @ gallery_features.t()                     # [Q, G]
 
     same_id = ql.eq(gl)                                              # [Q, G]
     if query_camera_labels is not None and gallery_camera_labels is not None:
