@@ -27,8 +27,11 @@ REPO_ROOT = os.path.dirname(os.path.abspath(__file__))
 if REPO_ROOT not in sys.path:
     sys.path.insert(0, REPO_ROOT)
 
-# benchmark mode: no ckpt audit trail in the timed loop
+# benchmark mode: no ckpt audit trail in the timed loop; MIOpen auto-tuned
+# conv algos (see tools/utils.same_seeds)
 os.environ.setdefault("FLREID_DISABLE_CKPT", "1")
+os.environ.setdefault("FLREID_FAST_CONV", "1")
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
 import torch  # noqa: E402
 
@@ -48,6 +51,8 @@ def parse_args():
                    help="train images per identity per task")
     p.add_argument("--num-classes", type=int, default=8000)
     p.add_argument("--cpu", action="store_true", help="debug on CPU")
+    p.add_argument("--channels-last", action="store_true",
+                   help="NHWC weights (MIOpen channels-last conv path)")
     return p.parse_args()
 
 
@@ -125,6 +130,9 @@ def main():
              if ctx.owner_of(i) == ctx.rank]
     clients = parser_clients(exp, common, owned_indices=owned)
     by_name = {c.client_name: c for c in clients}
+    if args.channels_last and device_is_cuda:
+        for c in clients:
+            c.model.net.to(memory_format=torch.channels_last)
 
     def one_round(r):
         stage.process_one_round(r, server, by_name, client_names, exp, log)

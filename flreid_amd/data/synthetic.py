@@ -31,6 +31,10 @@ def _seed_from(*parts) -> int:
     return int.from_bytes(h[:8], "little") % (2 ** 31)
 
 
+# process-wide cache of materialised synthetic splits (see _materialized)
+_SPLIT_CACHE: Dict = {}
+
+
 def parse_synthetic_dir(datasets_dir: str) -> Optional[Dict]:
     if not datasets_dir.startswith("synthetic:"):
         return None
@@ -89,10 +93,27 @@ class SyntheticReIDDataset(Dataset):
         noise = torch.randn(self.shape, generator=gs) * 0.25
         return (pattern + noise).clamp_(-1.0, 2.0)
 
+    def _materialized(self) -> torch.Tensor:
+        """Whole-split tensor, generated once per (task, split, geometry) and
+        cached process-wide — the task repeats for `sustain_rounds` rounds, so
+        regenerating per round is pure host overhead."""
+        key = (self.task_name, self.split, self.n_ids, self.imgs_per_id,
+               self.shape, tuple(self.classes))
+        cached = _SPLIT_CACHE.get(key)
+        if cached is None:
+            imgs = [self._image(self.classes[i // self.imgs_per_id],
+                                i % self.imgs_per_id)
+                    for i in range(len(self))]
+            cached = torch.stack(imgs)
+            if len(_SPLIT_CACHE) > 64:     # bound memory across many tasks
+                _SPLIT_CACHE.clear()
+            _SPLIT_CACHE[key] = cached
+        return cached
+
     def __getitem__(self, index: int):
         class_index = index // self.imgs_per_id
         person_id = self.classes[class_index]
-        img = self._image(person_id, index % self.imgs_per_id)
+        img = self._materialized()[index]
         if self.transform is not None:
             img = self.transform(img)
         return img, person_id, class_index

@@ -35,10 +35,9 @@ import math
 from typing import Any, Dict, List, Optional
 
 import torch
-from torch.utils.data import ConcatDataset, DataLoader
+from torch.utils.data import DataLoader
 
 from flreid_amd import ops
-from flreid_amd.data.loader import ReIDImageDataset
 from flreid_amd.methods.common import BaseReIDClient, BaseReIDOperator
 from flreid_amd.models.adaptive import (
     adaptive_leaves,
@@ -103,11 +102,25 @@ class Model(ModelModule):
     def m(self) -> int:
         return math.ceil(self.lambda_k / max(1, len(self.ids)))
 
+    def examplar_tensors(self, device):
+        """Stacked (data, pids, classes) of the exemplar store; data lives on
+        `device` (HBM-resident across rounds — 288 GB budget)."""
+        if not self.examplars:
+            return None, None, None
+        datas, pids, classes = [], [], []
+        for pid, entry in self.examplars.items():
+            protos_t, classes_t = entry
+            datas.append(protos_t.to(device))
+            pids.append(torch.full((protos_t.shape[0],), int(pid), dtype=torch.long))
+            classes.append(classes_t)
+        return torch.cat(datas), torch.cat(pids), torch.cat(classes)
+
     @torch.no_grad()
-    def build_examplars(self, proto_loader: DataLoader, person_ids, device) -> None:
+    def build_examplars(self, proto_loader, person_ids, device) -> None:
         """Herding selection per identity (ref:methods/fedstil.py:353-399):
         iteratively pick argmin‖μ − (f + Σ picked)/(i+1)‖ (repeats allowed,
-        matching the reference's selection loop)."""
+        matching the reference's selection loop).  Runs on-device; exemplars
+        stay in HBM."""
         protos, pids, classes, feats = [], [], [], []
         self.eval()
         for data, person_id, class_id in proto_loader:
@@ -116,10 +129,10 @@ class Model(ModelModule):
                 _score_feat = self.head_forward(data)
             # train-mode tuple or eval feature — capture the feature part
             feature = _score_feat[1] if isinstance(_score_feat, tuple) else _score_feat
-            protos.append(data.cpu())
+            protos.append(data)
             pids.append(person_id)
             classes.append(class_id)
-            feats.append(feature.float().cpu())
+            feats.append(feature.float())
         if not protos:
             return
         protos = torch.cat(protos)
@@ -130,26 +143,30 @@ class Model(ModelModule):
         if person_ids is not None and len(person_ids):
             keep = torch.tensor([int(p) in set(int(x) for x in person_ids)
                                  for p in pids], dtype=torch.bool)
-            protos, pids, classes, feats = protos[keep], pids[keep], classes[keep], feats[keep]
+            protos, pids, classes, feats = protos[keep.to(protos.device)], \
+                pids[keep], classes[keep], feats[keep.to(feats.device)]
 
         for person in torch.unique(pids).tolist():
             sel = (pids == person)
-            f = feats[sel]                    # [n, D]
-            p = protos[sel]
+            f = feats[sel.to(feats.device)]   # [n, D]
+            p = protos[sel.to(protos.device)]
             c = classes[sel]
             mu = f.mean(dim=0)
-            picked: List = []
+            idxs = []
             acc = torch.zeros_like(mu)
             for i in range(self.m):
                 cand = mu - (f + acc) / (i + 1)
                 idx = int(torch.linalg.vector_norm(cand, dim=1).argmin())
-                picked.append((p[idx].clone(), int(c[idx])))
+                idxs.append(idx)
                 acc = acc + f[idx]
-            self.examplars[int(person)] = picked
+            sel_idx = torch.tensor(idxs, dtype=torch.long, device=p.device)
+            self.examplars[int(person)] = (p.index_select(0, sel_idx).clone(),
+                                           c.index_select(0, sel_idx.cpu()).clone())
 
     def reduce_examplars(self) -> None:
-        for k in self.examplars:
-            self.examplars[k] = self.examplars[k][:self.m]
+        for k in list(self.examplars):
+            protos_t, classes_t = self.examplars[k]
+            self.examplars[k] = (protos_t[:self.m], classes_t[:self.m])
 
     # ------------------------------------------------------------ state I/O
     def model_state(self) -> Dict:
@@ -200,11 +217,45 @@ class Model(ModelModule):
         return ops.l1_drift(pairs)
 
 
+class TensorBatches:
+    """Device-resident rehearsal batcher.
+
+    Replaces the reference's per-item ConcatDataset + DataLoader over python
+    lists (ref:methods/fedstil.py:649-663): prototypes and exemplars stay in
+    HBM (288 GB budget — SURVEY.md §2.9 K10), batching is a randperm +
+    narrow, zero host round-trips.
+    """
+
+    def __init__(self, data: torch.Tensor, pids: torch.Tensor,
+                 classes: torch.Tensor, batch_size: int, shuffle: bool = True):
+        self.data, self.pids, self.classes = data, pids, classes
+        self.batch_size = batch_size
+        self.shuffle = shuffle
+        n = data.shape[0]
+        self.drop_last = n % batch_size == 1
+        self._len = n // batch_size + (0 if (self.drop_last or n % batch_size == 0) else 1)
+
+    def __len__(self):
+        return self._len
+
+    def __iter__(self):
+        n = self.data.shape[0]
+        order = torch.randperm(n) if self.shuffle else torch.arange(n)
+        order = order.to(self.data.device)
+        stop = self._len * self.batch_size if self.drop_last else n
+        for i in range(0, stop, self.batch_size):
+            idx = order[i:i + self.batch_size]
+            yield (self.data.index_select(0, idx),
+                   self.pids.index_select(0, idx.cpu()),
+                   self.classes.index_select(0, idx.cpu()))
+
+
 class Operator(BaseReIDOperator):
     def generate_proto_loader(self, model: Model, source_loader: DataLoader):
         """Prototype capture pass (ref:methods/fedstil.py:558-617): one frozen
         full-backbone forward over the task loader; yields the rehearsal
-        loader (exemplars ∪ current protos) and the task token."""
+        batcher (exemplars ∪ current protos, device-resident) and the task
+        token."""
         device = model.device
         taps, pids, classes = [], [], []
         model.eval()
@@ -213,28 +264,24 @@ class Operator(BaseReIDOperator):
                 data = data.to(device, non_blocking=True)
                 with autocast(device):
                     _out, tap = model.tap_forward(data)
-                taps.append(tap.float().cpu())
+                taps.append(tap.float())
                 pids.append(person_id)
                 classes.append(class_id)
-        taps = torch.cat(taps)
+        taps = torch.cat(taps)                      # stays on device
         pids = torch.cat(pids)
         classes = torch.cat(classes)
 
-        protos: Dict[int, List] = {}
-        for i in range(len(taps)):
-            protos.setdefault(int(pids[i]), []).append((taps[i], int(classes[i])))
+        ex_data, ex_pids, ex_classes = model.examplar_tensors(device)
+        if ex_data is not None:
+            all_data = torch.cat([ex_data, taps])
+            all_pids = torch.cat([ex_pids, pids])
+            all_classes = torch.cat([ex_classes, classes])
+        else:
+            all_data, all_pids, all_classes = taps, pids, classes
 
-        dataset = ConcatDataset([
-            ReIDImageDataset(source=model.examplars),
-            ReIDImageDataset(source=protos),
-        ])
-        loader = DataLoader(
-            dataset=dataset, shuffle=True,
-            batch_size=source_loader.batch_size,
-            num_workers=0,
-            drop_last=len(dataset) % source_loader.batch_size == 1,
-        )
-        task_token = taps.view(taps.shape[0], -1).mean(dim=0)
+        loader = TensorBatches(all_data, all_pids, all_classes,
+                               source_loader.batch_size, shuffle=True)
+        task_token = taps.view(taps.shape[0], -1).mean(dim=0).cpu()
         return loader, task_token
 
     def invoke_train(self, model: Model, dataloader: DataLoader, **kwargs) -> Any:
@@ -285,16 +332,41 @@ class Client(BaseReIDClient):
         self.task_token: Optional[torch.Tensor] = None
 
     # model ckpts use the six-section schema + a separate exemplar ckpt
-    # (ref:methods/fedstil.py:833-846)
+    # (ref:methods/fedstil.py:833-846); the ckpt keeps the reference's
+    # {pid: [(proto, class), ...]} layout on CPU while the live store is
+    # device-resident stacked tensors
+    @staticmethod
+    def _examplars_to_ckpt(examplars: Dict) -> Dict:
+        out = {}
+        for pid, (protos_t, classes_t) in examplars.items():
+            out[pid] = [(protos_t[i].cpu(), int(classes_t[i]))
+                        for i in range(protos_t.shape[0])]
+        return out
+
+    @staticmethod
+    def _examplars_from_ckpt(ck: Dict) -> Dict:
+        out = {}
+        for pid, items in ck.items():
+            if isinstance(items, tuple):      # already tensorised
+                out[pid] = items
+                continue
+            protos = torch.stack([p for p, _c in items]) if items else torch.empty(0)
+            classes = torch.tensor([c for _p, c in items], dtype=torch.long)
+            out[pid] = (protos, classes)
+        return out
+
     def load_model(self, model_name: str) -> None:
         model_dict = self.model.model_state()
         model_dict = self.load_state(model_name, model_dict)
         self.model.update_model(model_dict)
-        self.model.examplars = self.load_state(f"{model_name}_examplars", {})
+        loaded = self.load_state(f"{model_name}_examplars", {"__empty__": True})
+        if "__empty__" not in loaded:
+            self.model.examplars = self._examplars_from_ckpt(loaded)
 
     def save_model(self, model_name: str) -> None:
         self.save_state(model_name, self.model.model_state(), True)
-        self.save_state(f"{model_name}_examplars", self.model.examplars, True)
+        self.save_state(f"{model_name}_examplars",
+                        self._examplars_to_ckpt(self.model.examplars), True)
 
     def update_model(self, params_state: Dict) -> None:
         self.model.update_model(params_state)
