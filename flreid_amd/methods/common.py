@@ -53,6 +53,7 @@ class BaseReIDOperator(OperatorModule):
         device = model.device
 
         model.train()
+        acc_dev = loss_dev = None     # device-side metric accumulators
         for data, person_id, _classes_id in dataloader:
             data = data.to(device, non_blocking=True)
             target = person_id.to(device, non_blocking=True)
@@ -66,10 +67,15 @@ class BaseReIDOperator(OperatorModule):
             loss.backward()
             self.optimizer.step()
             score = output["score"]
-            train_acc += (score.detach().argmax(dim=1) == target).sum().item()
-            train_loss += float(loss.detach())
+            b_acc = (score.detach().argmax(dim=1) == target).sum()
+            b_loss = loss.detach()
+            acc_dev = b_acc if acc_dev is None else acc_dev + b_acc
+            loss_dev = b_loss if loss_dev is None else loss_dev + b_loss
             data_cnt += len(data)
             batch_cnt += 1
+        if acc_dev is not None:       # single host sync per epoch
+            train_acc = float(acc_dev)
+            train_loss = float(loss_dev)
 
         train_acc /= max(1, data_cnt)
         train_loss /= max(1, batch_cnt)

@@ -152,14 +152,16 @@ class Model(ModelModule):
             p = protos[sel.to(protos.device)]
             c = classes[sel]
             mu = f.mean(dim=0)
+            # sync-free herding: indices stay device tensors until one final
+            # host transfer per identity
             idxs = []
             acc = torch.zeros_like(mu)
             for i in range(self.m):
                 cand = mu - (f + acc) / (i + 1)
-                idx = int(torch.linalg.vector_norm(cand, dim=1).argmin())
+                idx = torch.linalg.vector_norm(cand, dim=1).argmin()
                 idxs.append(idx)
-                acc = acc + f[idx]
-            sel_idx = torch.tensor(idxs, dtype=torch.long, device=p.device)
+                acc = acc + f.index_select(0, idx.reshape(1)).squeeze(0)
+            sel_idx = torch.stack(idxs).to(p.device)
             self.examplars[int(person)] = (p.index_select(0, sel_idx).clone(),
                                            c.index_select(0, sel_idx.cpu()).clone())
 
@@ -240,14 +242,15 @@ class TensorBatches:
 
     def __iter__(self):
         n = self.data.shape[0]
-        order = torch.randperm(n) if self.shuffle else torch.arange(n)
-        order = order.to(self.data.device)
+        order_cpu = torch.randperm(n) if self.shuffle else torch.arange(n)
+        order_dev = order_cpu.to(self.data.device, non_blocking=True)
         stop = self._len * self.batch_size if self.drop_last else n
         for i in range(0, stop, self.batch_size):
-            idx = order[i:i + self.batch_size]
-            yield (self.data.index_select(0, idx),
-                   self.pids.index_select(0, idx.cpu()),
-                   self.classes.index_select(0, idx.cpu()))
+            idx_cpu = order_cpu[i:i + self.batch_size]
+            idx_dev = order_dev[i:i + self.batch_size]
+            yield (self.data.index_select(0, idx_dev),
+                   self.pids.index_select(0, idx_cpu),
+                   self.classes.index_select(0, idx_cpu))
 
 
 class Operator(BaseReIDOperator):
@@ -281,7 +284,7 @@ class Operator(BaseReIDOperator):
 
         loader = TensorBatches(all_data, all_pids, all_classes,
                                source_loader.batch_size, shuffle=True)
-        task_token = taps.view(taps.shape[0], -1).mean(dim=0).cpu()
+        task_token = taps.reshape(taps.shape[0], -1).mean(dim=0).cpu()
         return loader, task_token
 
     def invoke_train(self, model: Model, dataloader: DataLoader, **kwargs) -> Any:
@@ -292,6 +295,7 @@ class Operator(BaseReIDOperator):
 
         model.train()
         self.set_optimizer_parameters(model)
+        acc_dev = loss_dev = None     # device-side metric accumulators
         for data, person_id, _class_id in proto_loader:
             data = data.to(device, non_blocking=True)
             target = person_id.to(device, non_blocking=True)
@@ -304,10 +308,15 @@ class Operator(BaseReIDOperator):
                 loss = loss + model.drift_loss() * model.lambda_l1
             loss.backward()
             self.optimizer.step()
-            train_acc += (score.detach().argmax(dim=1) == target).sum().item()
-            train_loss += float(loss.detach())
+            b_acc = (score.detach().argmax(dim=1) == target).sum()
+            b_loss = loss.detach()
+            acc_dev = b_acc if acc_dev is None else acc_dev + b_acc
+            loss_dev = b_loss if loss_dev is None else loss_dev + b_loss
             data_cnt += len(data)
             batch_cnt += 1
+        if acc_dev is not None:       # single host sync per epoch
+            train_acc = float(acc_dev)
+            train_loss = float(loss_dev)
 
         if self.scheduler:
             self.scheduler.step()

@@ -127,17 +127,26 @@ class ExperimentStage:
         val_interval = int(exp_config["exp_opts"]["val_interval"])
         persist_comm = bool(exp_config.get("exp_opts", {}).get("persist_comm_ckpts", True))
 
-        # ---- dispatch (server -> clients); replicated computation ----------
+        # ---- dispatch (server -> clients); replicated server state ---------
+        # every rank registers every online client (keeps the replicated
+        # server deterministic), but the potentially expensive dispatch state
+        # (e.g. FedSTIL's personalized mixture) is only computed where it is
+        # consumed: on the owning rank, or on rank 0 for the ckpt audit trail
         for cname in online:
-            if cname not in server.clients:
+            first_contact = cname not in server.clients
+            if first_contact:
                 server.register_client(cname)
-                dispatch_state = server.get_dispatch_integrated_state(cname)
-                if dispatch_state is not None and cname in by_name:
-                    by_name[cname].update_by_integrated_state(dispatch_state)
-            else:
-                dispatch_state = server.get_dispatch_incremental_state(cname)
-                if dispatch_state is not None and cname in by_name:
-                    by_name[cname].update_by_incremental_state(dispatch_state)
+            needed = (cname in by_name) or (persist_comm and self.ctx.is_rank0())
+            dispatch_state = None
+            if needed:
+                if first_contact:
+                    dispatch_state = server.get_dispatch_integrated_state(cname)
+                    if dispatch_state is not None and cname in by_name:
+                        by_name[cname].update_by_integrated_state(dispatch_state)
+                else:
+                    dispatch_state = server.get_dispatch_incremental_state(cname)
+                    if dispatch_state is not None and cname in by_name:
+                        by_name[cname].update_by_incremental_state(dispatch_state)
             if persist_comm:
                 server.save_state(f"{curr_round}-{server.server_name}-{cname}",
                                   dispatch_state, True)
