@@ -146,24 +146,44 @@ class Model(ModelModule):
             protos, pids, classes, feats = protos[keep.to(protos.device)], \
                 pids[keep], classes[keep], feats[keep.to(feats.device)]
 
-        for person in torch.unique(pids).tolist():
-            sel = (pids == person)
-            f = feats[sel.to(feats.device)]   # [n, D]
-            p = protos[sel.to(protos.device)]
-            c = classes[sel]
-            mu = f.mean(dim=0)
-            # sync-free herding: indices stay device tensors until one final
-            # host transfer per identity
-            idxs = []
-            acc = torch.zeros_like(mu)
-            for i in range(self.m):
-                cand = mu - (f + acc) / (i + 1)
-                idx = torch.linalg.vector_norm(cand, dim=1).argmin()
-                idxs.append(idx)
-                acc = acc + f.index_select(0, idx.reshape(1)).squeeze(0)
-            sel_idx = torch.stack(idxs).to(p.device)
-            self.examplars[int(person)] = (p.index_select(0, sel_idx).clone(),
-                                           c.index_select(0, sel_idx.cpu()).clone())
+        # batched sync-free herding: all identities advance together, padded
+        # to the largest per-identity sample count; 4 kernels per herding step
+        # instead of 4 × n_identities (measured 8k+ micro-launches/round on
+        # the per-identity loop)
+        persons = torch.unique(pids).tolist()
+        per_idx = {p: (pids == p).nonzero(as_tuple=True)[0] for p in persons}
+        nmax = max(ix.numel() for ix in per_idx.values())
+        P = len(persons)
+        dev = feats.device
+        D = feats.shape[1]
+        f_pad = torch.zeros(P, nmax, D, device=dev)
+        pad_penalty = torch.full((P, nmax), 0.0, device=dev)
+        for r, p in enumerate(persons):
+            ix = per_idx[p].to(dev)
+            f_pad[r, :ix.numel()] = feats.index_select(0, ix)
+            if ix.numel() < nmax:
+                pad_penalty[r, ix.numel():] = 1e30
+        counts = torch.tensor([per_idx[p].numel() for p in persons],
+                              device=dev, dtype=torch.float32)
+        mu = f_pad.sum(dim=1) / counts.unsqueeze(1)              # [P, D]
+
+        acc = torch.zeros(P, D, device=dev)
+        arange_p = torch.arange(P, device=dev)
+        step_idx = []
+        for i in range(self.m):
+            cand = mu.unsqueeze(1) - (f_pad + acc.unsqueeze(1)) / (i + 1)
+            norms = torch.linalg.vector_norm(cand, dim=2) + pad_penalty
+            idx = norms.argmin(dim=1)                            # [P]
+            step_idx.append(idx)
+            acc = acc + f_pad[arange_p, idx]
+        sel = torch.stack(step_idx, dim=1).cpu()                 # [P, m]
+
+        for r, p in enumerate(persons):
+            local = per_idx[p]                                   # cpu indices
+            chosen = local[sel[r]]                               # [m] cpu
+            self.examplars[int(p)] = (
+                protos.index_select(0, chosen.to(protos.device)).clone(),
+                classes.index_select(0, chosen).clone())
 
     def reduce_examplars(self) -> None:
         for k in list(self.examplars):
@@ -216,7 +236,7 @@ class Model(ModelModule):
         pairs = []
         for _n, l in self.adaptive_module_leaves():
             pairs.extend(l.drift_pairs())
-        return ops.l1_drift(pairs)
+        return ops.l1_drift_fused(pairs)
 
 
 class TensorBatches:

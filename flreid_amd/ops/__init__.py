@@ -243,6 +243,7 @@ kl_distance = ref.kl_distance
 kd_loss = ref.kd_loss
 quadratic_penalty = ref.quadratic_penalty
 l1_drift = ref.l1_drift
+l1_drift_fused = ref.l1_drift_fused
 
 
 def triplet_loss(feature: torch.Tensor, target: torch.Tensor,
