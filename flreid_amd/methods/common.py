@@ -22,16 +22,42 @@ from flreid_amd.tools.evaluate import calculate_similarity_distance, evaluate
 from flreid_amd.tools.utils import model_on_device
 
 
+def reset_optimizer_state_inplace(optimizer) -> None:
+    """Semantically identical to the reference's between-task state drop
+    (`optimizer.state = defaultdict(dict)`, ref:methods/fedavg.py:306-309)
+    but zeroes the existing state tensors IN PLACE so captured hipGraphs
+    (runtime/hipgraph.py) stay valid across rounds."""
+    import collections
+
+    for state in optimizer.state.values():
+        zeroed = True
+        for k, v in state.items():
+            if torch.is_tensor(v):
+                v.zero_()
+            elif isinstance(v, (int, float)):
+                zeroed = False
+        if not zeroed:
+            # non-tensor step counters can't be zeroed in place safely across
+            # torch versions — fall back to a full drop for this optimizer
+            optimizer.state = collections.defaultdict(dict)
+            return
+
+
 class BaseReIDOperator(OperatorModule):
     """Per-epoch train / predict / valid / inference loops
     (ref:methods/fedavg.py:27-211)."""
 
-    def set_optimizer_parameters(self, model) -> None:
+    def set_optimizer_parameters(self, model, capturable: bool = False) -> None:
         """Rebind the optimizer to the model's current requires_grad set
         (needed after dispatch re-init — ref:methods/fedstil.py:552-555)."""
         defaults = dict(self.optimizer.defaults)
+        if capturable and "capturable" in defaults:
+            defaults["capturable"] = True
         params = [p for p in model.net.parameters() if p.requires_grad]
-        self.optimizer.param_groups = [{"params": params, **defaults}]
+        group = {"params": params, **defaults}
+        if capturable:
+            group["capturable"] = True
+        self.optimizer.param_groups = [group]
 
     # hook: extra loss terms (EWC/MAS penalty, FedProx prox, FedSTIL L1 ...)
     def penalty(self, model) -> Optional[torch.Tensor]:
@@ -191,7 +217,7 @@ class BaseReIDClient(ClientModule):
             self.after_task_train(output, tr_loader, device)
 
         # reset optimizer state + LR between tasks (ref:methods/fedavg.py:306-309)
-        self.operator.optimizer.state = collections.defaultdict(dict)
+        reset_optimizer_state_inplace(self.operator.optimizer)
         for group in self.operator.optimizer.param_groups:
             group["lr"] = initial_lr
 

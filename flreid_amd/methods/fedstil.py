@@ -307,31 +307,69 @@ class Operator(BaseReIDOperator):
         task_token = taps.reshape(taps.shape[0], -1).mean(dim=0).cpu()
         return loader, task_token
 
+    def _train_step(self, model: Model, data: torch.Tensor,
+                    target: torch.Tensor):
+        """One head-training step — eager body AND the hipGraph-captured fn
+        (must stay replay-safe: fixed shapes, no host syncs)."""
+        device = model.device
+        self.optimizer.zero_grad(set_to_none=False)
+        with autocast(device):
+            score, feature = model.head_forward(data)
+            loss = 0.0
+            for loss_func in self.criterion:
+                loss = loss + loss_func(score=score, feature=feature, target=target)
+            loss = loss + model.drift_loss() * model.lambda_l1
+        loss.backward()
+        self.optimizer.step()
+        b_acc = (score.detach().argmax(dim=1) == target).sum()
+        return loss.detach(), b_acc
+
+    def _graphed_step(self, model: Model, batch_size: int):
+        """Per-(operator, batch-size, lr) cached GraphedStep; recaptured when
+        the LR schedule moves (the captured Adam step bakes the lr)."""
+        from flreid_amd.runtime.hipgraph import GraphedStep
+
+        lr = self.optimizer.param_groups[0]["lr"]
+        key = (batch_size, float(lr))
+        cache = getattr(self, "_train_graphs", None)
+        if cache is None or cache[0] != key:
+            gs = GraphedStep(lambda d, t: self._train_step(model, d, t))
+            self._train_graphs = (key, gs)
+        return self._train_graphs[1]
+
     def invoke_train(self, model: Model, dataloader: DataLoader, **kwargs) -> Any:
+        from flreid_amd.runtime.hipgraph import hipgraph_enabled
+
         train_acc = train_loss = 0.0
         batch_cnt = data_cnt = 0
         device = model.device
+        use_graph = hipgraph_enabled() and str(device).startswith("cuda")
         proto_loader, task_token = self.generate_proto_loader(model, dataloader)
 
         model.train()
-        self.set_optimizer_parameters(model)
-        acc_dev = loss_dev = None     # device-side metric accumulators
+        self.set_optimizer_parameters(model, capturable=use_graph)
+        batch_size = getattr(proto_loader, "batch_size", None)
+        gs = self._graphed_step(model, batch_size) if use_graph else None
+        warmups = 0
+
+        # device-side metric accumulators; always out-of-place adds so replay
+        # outputs (static tensors) are consumed before the next replay
+        acc_dev = loss_dev = None
         for data, person_id, _class_id in proto_loader:
             data = data.to(device, non_blocking=True)
             target = person_id.to(device, non_blocking=True)
-            self.optimizer.zero_grad(set_to_none=True)
-            with autocast(device):
-                score, feature = model.head_forward(data)
-                loss = 0.0
-                for loss_func in self.criterion:
-                    loss = loss + loss_func(score=score, feature=feature, target=target)
-                loss = loss + model.drift_loss() * model.lambda_l1
-            loss.backward()
-            self.optimizer.step()
-            b_acc = (score.detach().argmax(dim=1) == target).sum()
-            b_loss = loss.detach()
-            acc_dev = b_acc if acc_dev is None else acc_dev + b_acc
-            loss_dev = b_loss if loss_dev is None else loss_dev + b_loss
+            if gs is not None and data.shape[0] == batch_size:
+                if gs.ready:
+                    b_loss, b_acc = gs(data, target)
+                elif warmups < 2:
+                    b_loss, b_acc = gs.warmup(data, target)
+                    warmups += 1
+                else:
+                    b_loss, b_acc = gs.capture(data, target)
+            else:
+                b_loss, b_acc = self._train_step(model, data, target)
+            acc_dev = b_acc.clone() if acc_dev is None else acc_dev + b_acc
+            loss_dev = b_loss.clone() if loss_dev is None else loss_dev + b_loss
             data_cnt += len(data)
             batch_cnt += 1
         if acc_dev is not None:       # single host sync per epoch
@@ -470,7 +508,8 @@ class Client(BaseReIDClient):
             self.model.build_examplars(output["proto_loader"],
                                        tr_loader.dataset.person_ids, device)
 
-        self.operator.optimizer.state = collections.defaultdict(dict)
+        from flreid_amd.methods.common import reset_optimizer_state_inplace
+        reset_optimizer_state_inplace(self.operator.optimizer)
         for group in self.operator.optimizer.param_groups:
             group["lr"] = initial_lr
 

@@ -53,30 +53,39 @@ class AdaptiveBase(nn.Module):
         self.init_training_weights(global_weight, global_weight_atten,
                                    adaptive_weight, adaptive_bias)
 
+    @staticmethod
+    def _assign(param: Parameter, value: torch.Tensor) -> None:
+        """In-place when shapes match (keeps optimizer state and captured
+        hipGraph pointers valid across per-round re-inits); rebind otherwise."""
+        if param.data.shape == value.shape:
+            param.data.copy_(value)
+        else:
+            param.data = value.detach().clone().to(param.device)
+
     @torch.no_grad()
     def init_training_weights(self, global_weight=None, global_weight_atten=None,
                               adaptive_weight=None, adaptive_bias=None) -> None:
         if global_weight is None:
             global_weight = self.global_weight.data
-        self.global_weight.data = global_weight.detach().clone()
+        self._assign(self.global_weight, global_weight.detach())
         self.global_weight.requires_grad = False
 
         if global_weight_atten is None:
             global_weight_atten = torch.ones(
                 self.global_weight.data.shape[-1],
                 device=self.global_weight.device) * self.atten_default
-        self.global_weight_atten.data = global_weight_atten.detach().clone()
-        self.initial_global_weight_atten.data = global_weight_atten.detach().clone()
+        self._assign(self.global_weight_atten, global_weight_atten.detach())
+        self._assign(self.initial_global_weight_atten, global_weight_atten.detach())
         self.global_weight_atten.requires_grad = self.atten_trainable
 
         if adaptive_weight is None:
             adaptive_weight = (1.0 - self.global_weight_atten.data) * self.global_weight.data
-        self.adaptive_weight.data = adaptive_weight.detach().clone()
-        self.initial_adaptive_weight.data = adaptive_weight.detach().clone()
+        self._assign(self.adaptive_weight, adaptive_weight.detach())
+        self._assign(self.initial_adaptive_weight, adaptive_weight.detach())
         self.adaptive_weight.requires_grad = True
 
         if self.adaptive_bias is not None and adaptive_bias is not None:
-            self.adaptive_bias.data = adaptive_bias.detach().clone()
+            self._assign(self.adaptive_bias, adaptive_bias.detach())
             self.adaptive_bias.requires_grad = True
 
     def composed_weight(self) -> torch.Tensor:
