@@ -344,11 +344,14 @@ class Operator(BaseReIDOperator):
         batch_cnt = data_cnt = 0
         device = model.device
         use_graph = hipgraph_enabled() and str(device).startswith("cuda")
-        proto_loader, task_token = self.generate_proto_loader(model, dataloader)
+        from flreid_amd.runtime.hipgraph import phase
+        with phase("proto_capture"):
+            proto_loader, task_token = self.generate_proto_loader(model, dataloader)
 
         model.train()
         self.set_optimizer_parameters(model, capturable=use_graph)
         batch_size = getattr(proto_loader, "batch_size", None)
+        _phase_head = phase("head_epoch"); _phase_head.__enter__()
         gs = self._graphed_step(model, batch_size) if use_graph else None
         warmups = 0
 
@@ -372,6 +375,7 @@ class Operator(BaseReIDOperator):
             loss_dev = b_loss.clone() if loss_dev is None else loss_dev + b_loss
             data_cnt += len(data)
             batch_cnt += 1
+        _phase_head.__exit__(None, None, None)
         if acc_dev is not None:       # single host sync per epoch
             train_acc = float(acc_dev)
             train_loss = float(loss_dev)
@@ -504,9 +508,11 @@ class Client(BaseReIDClient):
                 self.logger.info_train(task_name, device, output["data_count"],
                                        perf_acc, perf_loss, epoch, epochs)
 
+            from flreid_amd.runtime.hipgraph import phase
             self.model.reduce_examplars()
-            self.model.build_examplars(output["proto_loader"],
-                                       tr_loader.dataset.person_ids, device)
+            with phase("herding"):
+                self.model.build_examplars(output["proto_loader"],
+                                           tr_loader.dataset.person_ids, device)
 
         from flreid_amd.methods.common import reset_optimizer_state_inplace
         reset_optimizer_state_inplace(self.operator.optimizer)

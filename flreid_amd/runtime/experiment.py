@@ -182,7 +182,9 @@ class ExperimentStage:
                 local_uploads[cname] = _state_to_cpu(state)
 
         # ---- sync uploads across ranks (ONE gather per round) --------------
-        gathered = self.ctx.all_gather_object(local_uploads)
+        from flreid_amd.runtime.hipgraph import phase as _phase
+        with _phase("upload_sync"):
+            gathered = self.ctx.all_gather_object(local_uploads)
         merged: Dict[str, Any] = {}
         for rank_uploads in gathered:
             merged.update(rank_uploads)
@@ -191,8 +193,12 @@ class ExperimentStage:
                 server.set_client_incremental_state(cname, merged[cname])
 
         # ---- aggregate (replicated, deterministic) -------------------------
-        server.calculate()
+        from flreid_amd.runtime.hipgraph import dump_phases, phase, phase_timers_enabled
+        with phase("aggregate"):
+            server.calculate()
         log.sync(self.ctx)
+        if phase_timers_enabled():
+            print(f"[phases r{curr_round}] {dump_phases()}", flush=True)
 
     # -------------------------------------------------------------- workers
     @clear_cache

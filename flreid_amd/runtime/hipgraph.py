@@ -73,3 +73,42 @@ class GraphedStep:
             dst.copy_(src, non_blocking=True)
         self.graph.replay()
         return self.static_outputs
+
+
+# ---------------------------------------------------------------------------
+# lightweight phase timing (FLREID_PHASE_TIMERS=1): per-phase wall clock with
+# a device sync at each boundary — for finding where a round's time goes
+# ---------------------------------------------------------------------------
+
+import contextlib
+import time
+from collections import defaultdict
+
+PHASE_TIMES = defaultdict(float)
+
+
+def phase_timers_enabled() -> bool:
+    return os.environ.get("FLREID_PHASE_TIMERS", "0") == "1"
+
+
+@contextlib.contextmanager
+def phase(name: str):
+    if not phase_timers_enabled():
+        yield
+        return
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    try:
+        yield
+    finally:
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        PHASE_TIMES[name] += time.perf_counter() - t0
+
+
+def dump_phases(reset: bool = True) -> str:
+    out = " ".join(f"{k}={v*1000:.0f}ms" for k, v in sorted(PHASE_TIMES.items()))
+    if reset:
+        PHASE_TIMES.clear()
+    return out
