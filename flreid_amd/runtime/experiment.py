@@ -136,9 +136,11 @@ class ExperimentStage:
             first_contact = cname not in server.clients
             if first_contact:
                 server.register_client(cname)
+            from flreid_amd.runtime.hipgraph import phase as _ph
             needed = (cname in by_name) or (persist_comm and self.ctx.is_rank0())
             dispatch_state = None
             if needed:
+              with _ph("dispatch"):
                 if first_contact:
                     dispatch_state = server.get_dispatch_integrated_state(cname)
                     if dispatch_state is not None and cname in by_name:
@@ -161,7 +163,9 @@ class ExperimentStage:
             if cname in by_name:
                 same_seeds((base_seed * 1000003 + curr_round * 1009 +
                             client_names.index(cname)) % (2 ** 31))
-                self._process_train(by_name[cname], log, curr_round)
+                from flreid_amd.runtime.hipgraph import phase as _ph3
+                with _ph3("train_total"):
+                    self._process_train(by_name[cname], log, curr_round)
 
         # ---- validation every val_interval rounds --------------------------
         if val_interval and curr_round % val_interval == 0:
@@ -174,12 +178,15 @@ class ExperimentStage:
             if cname not in by_name:
                 continue
             client = by_name[cname]
-            state = client.get_incremental_state()
+            from flreid_amd.runtime.hipgraph import phase as _ph2
+            with _ph2("upload_build"):
+                state = client.get_incremental_state()
             if persist_comm:
                 client.save_state(f"{curr_round}-{cname}-{server.server_name}",
                                   state, True)
             if state is not None:
-                local_uploads[cname] = _state_to_cpu(state)
+                with _ph2("upload_tocpu"):
+                    local_uploads[cname] = _state_to_cpu(state)
 
         # ---- sync uploads across ranks (ONE gather per round) --------------
         from flreid_amd.runtime.hipgraph import phase as _phase

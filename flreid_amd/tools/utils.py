@@ -75,12 +75,22 @@ def model_on_device(model, device: str = "cpu"):
             model.cpu()
 
 
+def _cache_clearing_enabled() -> bool:
+    # The reference gc'd + emptied the HIP cache after every train/val job
+    # (ref:tools/utils.py:124-136) because thread-pooled clients shared GPUs.
+    # With one resident process per GPU that churns the allocator for nothing
+    # (measured tens of ms per round + reallocation stalls), so it is opt-in.
+    return os.environ.get("FLREID_EMPTY_CACHE", "0") == "1"
+
+
 def clear_cache(fn=None):
-    """Decorator: gc + empty HIP cache after the call (ref:tools/utils.py:124-136)."""
+    """Decorator: gc + empty HIP cache after the call (ref:tools/utils.py:124-136);
+    active only when FLREID_EMPTY_CACHE=1 (see _cache_clearing_enabled)."""
     if fn is None:
-        gc.collect()
-        if torch.cuda.is_available():
-            torch.cuda.empty_cache()
+        if _cache_clearing_enabled():
+            gc.collect()
+            if torch.cuda.is_available():
+                torch.cuda.empty_cache()
         return None
 
     @functools.wraps(fn)
@@ -88,9 +98,10 @@ def clear_cache(fn=None):
         try:
             return fn(*args, **kwargs)
         finally:
-            gc.collect()
-            if torch.cuda.is_available():
-                torch.cuda.empty_cache()
+            if _cache_clearing_enabled():
+                gc.collect()
+                if torch.cuda.is_available():
+                    torch.cuda.empty_cache()
 
     return wrapped
 
