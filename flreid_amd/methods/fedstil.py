@@ -215,16 +215,27 @@ class Model(ModelModule):
         return state
 
     def update_model(self, params_state: Dict) -> None:
-        merged: Dict[str, torch.Tensor] = {}
-        for section in ("global_weight", "global_weight_atten", "adaptive_weights",
-                        "adaptive_bias", "bn_params", "pre_trained_params"):
-            for n, p in params_state.get(section, {}).items():
-                merged[n] = p.detach().clone()
-        model_dict = self.net.state_dict()
-        for n, p in merged.items():
-            if n in model_dict:
-                model_dict[n] = p.to(model_dict[n].device, model_dict[n].dtype)
-        self.net.load_state_dict(model_dict)
+        """Targeted in-place copy of just the provided keys (the reference
+        re-loaded the ENTIRE state dict for a 10-tensor dispatch —
+        ref:methods/fedstil.py:535-547)."""
+        live = {n: t for n, t in self.net.state_dict(keep_vars=True).items()}
+        with torch.no_grad():
+            for section in ("global_weight", "global_weight_atten",
+                            "adaptive_weights", "adaptive_bias", "bn_params",
+                            "pre_trained_params"):
+                for n, p in params_state.get(section, {}).items():
+                    dst = live.get(n)
+                    if dst is None:
+                        continue
+                    if dst.shape == p.shape:
+                        dst.copy_(p.detach(), non_blocking=True)
+                    else:
+                        # shape change (e.g. the stacked-atten variant):
+                        # rebind the parameter's storage
+                        mod = self.net.get_submodule(n.rsplit(".", 1)[0])
+                        leaf = n.rsplit(".", 1)[1]
+                        getattr(mod, leaf).data = p.detach().clone().to(
+                            dst.device, dst.dtype)
 
     def composed_upload(self) -> Dict[str, torch.Tensor]:
         """{name.global_weight: atten⊙W_glob + W_adapt} — what the client
@@ -427,12 +438,14 @@ class Client(BaseReIDClient):
         return out
 
     def load_model(self, model_name: str) -> None:
-        model_dict = self.model.model_state()
-        model_dict = self.load_state(model_name, model_dict)
-        self.model.update_model(model_dict)
-        loaded = self.load_state(f"{model_name}_examplars", {"__empty__": True})
-        if "__empty__" not in loaded:
-            self.model.examplars = self._examplars_from_ckpt(loaded)
+        # re-applying the model's own current state is a no-op: skip the full
+        # clone + copy when no checkpoint exists (resident models make the
+        # reference's load-before-every-use redundant)
+        if self.state_exists(model_name):
+            self.model.update_model(self.load_state(model_name, None))
+        if self.state_exists(f"{model_name}_examplars"):
+            self.model.examplars = self._examplars_from_ckpt(
+                self.load_state(f"{model_name}_examplars", {}))
 
     def save_model(self, model_name: str) -> None:
         self.save_state(model_name, self.model.model_state(), True)
@@ -566,11 +579,7 @@ class Server(ServerModule):
             for n, p in s["incremental_sw"].items():
                 contrib = p.detach().to(torch.float32) * (k / total)
                 merged[n] = merged.get(n, 0) + contrib
-        model_dict = self.model.net.state_dict()
-        for n, p in merged.items():
-            if n in model_dict:
-                model_dict[n] = p.to(model_dict[n].dtype)
-        self.model.net.load_state_dict(model_dict)
+        self.model.update_model({"global_weight": merged})
         self.save_state(f"{self.server_name}_tokens", self.token_memory, True)
 
     def _remember_token(self, client_name: str, state: Dict) -> None:

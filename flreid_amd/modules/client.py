@@ -60,6 +60,11 @@ class ClientModule:
             return default_value
         raise ValueError(f"State checkpoint does not exist in '{state_path}'.")
 
+    def state_exists(self, state_name: str) -> bool:
+        if self._ckpt_disabled():
+            return False
+        return os.path.exists(os.path.join(self.ckpt_path, f"{state_name}.ckpt"))
+
     def save_state(self, state_name: str, state: Any, cover: bool = False) -> None:
         if state_name is None or self._ckpt_disabled():
             return

@@ -92,6 +92,11 @@ class ExperimentStage:
         self.logger.info(f"Experiment loading succeed: {exp_config['exp_name']}")
 
         server = parser_server(exp_config, self.common_config)
+        if str(self.device).startswith("cuda"):
+            # replicated server lives in HBM next to its rank's clients:
+            # dispatch mixtures / weighted averages run at HBM bandwidth
+            # instead of host memory (288 GB leaves plenty of room)
+            server.model.to(self.device)
         client_names = [c["client_name"] for c in exp_config["clients"]]
         owned = [i for i in range(len(client_names))
                  if self.ctx.owner_of(i) == self.ctx.rank]
@@ -185,8 +190,13 @@ class ExperimentStage:
                 client.save_state(f"{curr_round}-{cname}-{server.server_name}",
                                   state, True)
             if state is not None:
-                with _ph2("upload_tocpu"):
-                    local_uploads[cname] = _state_to_cpu(state)
+                if self.ctx.is_distributed:
+                    # object-gather path needs host tensors (a pickled CUDA
+                    # tensor would try to restore onto the sender's device)
+                    with _ph2("upload_tocpu"):
+                        local_uploads[cname] = _state_to_cpu(state)
+                else:
+                    local_uploads[cname] = state
 
         # ---- sync uploads across ranks (ONE gather per round) --------------
         from flreid_amd.runtime.hipgraph import phase as _phase
