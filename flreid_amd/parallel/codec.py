@@ -1,0 +1,144 @@
+"""Tensor codec for client-state synchronisation (C2/C3 in SURVEY.md §2.9).
+
+Upload states are nested python dicts of tensors + scalars.  The generic
+`all_gather_object` path pickles ~100 MB of parameters per client through
+host memory; this codec instead:
+
+  1. exchanges a tiny metadata skeleton (names, shapes, dtypes, scalars)
+     via object gather;
+  2. packs every client's tensors into ONE flat device buffer (sorted key
+     order), pads ranks to the same client count, and runs a single RCCL
+     `all_gather_into_tensor` over xGMI;
+  3. rebuilds per-client states as DEVICE tensors on every rank, so server
+     aggregation stays at HBM bandwidth.
+
+Falls back to the object path when client schemas differ across ranks (e.g.
+methods with data-dependent state shapes).
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Tuple
+
+import torch
+
+_SENTINEL = "__flreid_tensor__"
+
+
+def _flatten(state: Any, path: str, tensors: List[Tuple[str, torch.Tensor]]):
+    """Returns a skeleton with tensors replaced by (sentinel, shape, dtype)."""
+    if torch.is_tensor(state):
+        tensors.append((path, state))
+        return (_SENTINEL, tuple(state.shape), str(state.dtype))
+    if isinstance(state, dict):
+        return {k: _flatten(state[k], f"{path}.{k}", tensors)
+                for k in sorted(state.keys(), key=str)}
+    if isinstance(state, (list, tuple)):
+        out = [_flatten(v, f"{path}[{i}]", tensors) for i, v in enumerate(state)]
+        return tuple(out) if isinstance(state, tuple) else out
+    return state
+
+
+def _rebuild(skeleton: Any, chunks: List[torch.Tensor], pos: List[int]) -> Any:
+    if isinstance(skeleton, tuple) and len(skeleton) == 3 and skeleton[0] == _SENTINEL:
+        t = chunks[pos[0]]
+        pos[0] += 1
+        return t
+    if isinstance(skeleton, dict):
+        return {k: _rebuild(v, chunks, pos) for k, v in skeleton.items()}
+    if isinstance(skeleton, (list, tuple)):
+        out = [_rebuild(v, chunks, pos) for v in skeleton]
+        return out if isinstance(skeleton, list) else tuple(out)
+    return skeleton
+
+
+def _dtype_of(name: str) -> torch.dtype:
+    return getattr(torch, name.replace("torch.", ""))
+
+
+def sync_client_states(ctx, local_uploads: Dict[str, Any]) -> Dict[str, Any]:
+    """All ranks end with the union of every rank's {client: state}."""
+    if not ctx.is_distributed:
+        return dict(local_uploads)
+
+    device = ctx._comm_device()
+
+    # 1. flatten locally
+    metas: Dict[str, Any] = {}
+    flats: Dict[str, torch.Tensor] = {}
+    for cname in sorted(local_uploads.keys()):
+        tensors: List[Tuple[str, torch.Tensor]] = []
+        skeleton = _flatten(local_uploads[cname], cname, tensors)
+        if tensors:
+            flat = torch.cat([t.detach().reshape(-1).to(torch.float32)
+                              for _n, t in tensors]).to(device)
+        else:
+            flat = torch.zeros(0, dtype=torch.float32, device=device)
+        metas[cname] = skeleton
+        flats[cname] = flat
+
+    # 2. metadata hop (tiny): skeletons + per-client payload sizes
+    per_client_sizes = {c: f.numel() for c, f in flats.items()}
+    all_metas = ctx.all_gather_object((metas, per_client_sizes))
+
+    strides = {n for _m, sizes in all_metas for n in sizes.values()}
+    counts = [len(sizes) for _m, sizes in all_metas]
+
+    # equal-stride fast path: every client's payload has the same numel
+    if len(strides) == 1 and any(counts):
+        stride = next(iter(strides))
+        max_clients = max(counts)
+        local = torch.zeros(max_clients * stride, dtype=torch.float32, device=device)
+        for i, cname in enumerate(sorted(flats.keys())):
+            local[i * stride:(i + 1) * stride] = flats[cname]
+        gathered = ctx.all_gather_flat(local)          # [W, max_clients*stride]
+
+        out: Dict[str, Any] = {}
+        for rank, (rank_metas, rank_sizes) in enumerate(all_metas):
+            for i, cname in enumerate(sorted(rank_sizes.keys())):
+                flat = gathered[rank, i * stride:(i + 1) * stride]
+                out[cname] = _unflatten_state(rank_metas[cname], flat)
+        return out
+
+    # fallback: ragged schemas -> object path (host round trip)
+    gathered_obj = ctx.all_gather_object({c: _to_cpu(s) for c, s in local_uploads.items()})
+    merged: Dict[str, Any] = {}
+    for rank_uploads in gathered_obj:
+        merged.update(rank_uploads)
+    return merged
+
+
+def _unflatten_state(skeleton: Any, flat: torch.Tensor) -> Any:
+    chunks: List[torch.Tensor] = []
+    offset = 0
+
+    def walk(node):
+        nonlocal offset
+        if isinstance(node, tuple) and len(node) == 3 and node[0] == _SENTINEL:
+            shape, dtype = node[1], _dtype_of(node[2])
+            numel = 1
+            for s in shape:
+                numel *= s
+            chunks.append(flat[offset:offset + numel].view(shape).to(dtype))
+            offset += numel
+            return
+        if isinstance(node, dict):
+            for v in node.values():
+                walk(v)
+        elif isinstance(node, (list, tuple)):
+            for v in node:
+                walk(v)
+
+    walk(skeleton)
+    return _rebuild(skeleton, chunks, [0])
+
+
+def _to_cpu(state: Any) -> Any:
+    if torch.is_tensor(state):
+        return state.detach().cpu()
+    if isinstance(state, dict):
+        return {k: _to_cpu(v) for k, v in state.items()}
+    if isinstance(state, (list, tuple)):
+        out = [_to_cpu(v) for v in state]
+        return out if isinstance(state, list) else tuple(out)
+    return state

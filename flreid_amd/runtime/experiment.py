@@ -190,21 +190,15 @@ class ExperimentStage:
                 client.save_state(f"{curr_round}-{cname}-{server.server_name}",
                                   state, True)
             if state is not None:
-                if self.ctx.is_distributed:
-                    # object-gather path needs host tensors (a pickled CUDA
-                    # tensor would try to restore onto the sender's device)
-                    with _ph2("upload_tocpu"):
-                        local_uploads[cname] = _state_to_cpu(state)
-                else:
-                    local_uploads[cname] = state
+                # the tensor codec keeps device tensors device-resident;
+                # its ragged-schema fallback cpu-ifies internally
+                local_uploads[cname] = state
 
         # ---- sync uploads across ranks (ONE gather per round) --------------
+        from flreid_amd.parallel.codec import sync_client_states
         from flreid_amd.runtime.hipgraph import phase as _phase
         with _phase("upload_sync"):
-            gathered = self.ctx.all_gather_object(local_uploads)
-        merged: Dict[str, Any] = {}
-        for rank_uploads in gathered:
-            merged.update(rank_uploads)
+            merged = sync_client_states(self.ctx, local_uploads)
         for cname in online:  # deterministic application order
             if cname in merged:
                 server.set_client_incremental_state(cname, merged[cname])

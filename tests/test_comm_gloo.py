@@ -178,3 +178,36 @@ def test_distributed_equals_single_process(tmp_path):
     assert set(single_state) == set(dist_state)
     for n in single_state:
         assert torch.allclose(single_state[n], dist_state[n], atol=1e-6), n
+
+
+def _worker_codec(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.codec import sync_client_states
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        state = {
+            "train_cnt": 10 + rank,
+            "task_token": torch.full((4,), float(rank)),
+            "incremental_sw": {
+                "layer.w": torch.full((3, 2), float(rank + 1)),
+                "layer.b": torch.arange(2.0) + rank,
+            },
+        }
+        merged = sync_client_states(ctx, {f"client-{rank}": state})
+        assert set(merged) == {"client-0", "client-1"}
+        for r in (0, 1):
+            s = merged[f"client-{r}"]
+            assert s["train_cnt"] == 10 + r
+            assert torch.allclose(s["task_token"], torch.full((4,), float(r)))
+            assert torch.allclose(s["incremental_sw"]["layer.w"],
+                                  torch.full((3, 2), float(r + 1)))
+            assert torch.allclose(s["incremental_sw"]["layer.b"],
+                                  torch.arange(2.0) + r)
+    finally:
+        destroy_context()
+
+
+def test_tensor_codec_gather(tmp_path):
+    port = _free_port()
+    mp.spawn(_worker_codec, args=(2, port, str(tmp_path)), nprocs=2, join=True)
