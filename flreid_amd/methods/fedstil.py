@@ -63,8 +63,7 @@ class Model(ModelModule):
         self.atten_default = atten_default
         self.args = kwargs
 
-        convert_to_adaptive(self.net, atten_default,
-                            atten_trainable=self.atten_trainable)
+        self._convert_net()
 
         self.ids = set()
         self.examplars: Dict[int, List] = {}
@@ -75,6 +74,10 @@ class Model(ModelModule):
             raise ValueError("fedstil requires at least one trainable "
                              "Linear/Conv2d (check fine_tuning)")
         self.head_stage = min(self.net.stage_of(n) for n, _ in leaves)
+
+    def _convert_net(self) -> None:
+        convert_to_adaptive(self.net, self.atten_default,
+                            atten_trainable=self.atten_trainable)
 
     # ------------------------------------------------------------ structure
     def adaptive_module_leaves(self):
@@ -341,7 +344,11 @@ class Operator(BaseReIDOperator):
         from flreid_amd.runtime.hipgraph import GraphedStep
 
         lr = self.optimizer.param_groups[0]["lr"]
-        key = (batch_size, float(lr))
+        # shapes matter: a stacked-atten dispatch grows parameter shapes,
+        # which invalidates captured pointers
+        shapes = tuple(tuple(p.shape) for p in
+                       self.optimizer.param_groups[0]["params"])
+        key = (batch_size, float(lr), shapes)
         cache = getattr(self, "_train_graphs", None)
         if cache is None or cache[0] != key:
             gs = GraphedStep(lambda d, t: self._train_step(model, d, t))

@@ -196,3 +196,119 @@ def non_adaptive_leaves(net: nn.Module):
         if len(list(m.children())) == 0 and not isinstance(m, ADAPTIVE_TYPES) and n:
             out.append((n, m))
     return out
+
+
+# ---------------------------------------------------------------------------
+# FedSTIL-atten stacked variant (ref:methods/fedstil_atten.py)
+# ---------------------------------------------------------------------------
+
+class StackedAttenBase(nn.Module):
+    """θ = Σ_k atten_k · W_glob[..., k] + W_adapt.
+
+    The server CONCATENATES client uploads along a new last dim instead of
+    averaging, and the per-slot attention is LEARNABLE; `adaptive_weight`
+    persists across dispatch re-inits while `atten` is re-initialised to
+    `atten_default` sized to the current stack depth
+    (ref:methods/fedstil_atten.py:46-90)."""
+
+    def __init__(self, global_weight: torch.Tensor,
+                 adaptive_bias: Optional[torch.Tensor] = None,
+                 atten_default: float = 0.0, **kwargs):
+        super().__init__()
+        self.atten_default = atten_default
+        self.atten_trainable = True
+        self.global_weight = Parameter(torch.empty(0), requires_grad=False)
+        self.global_weight_atten = Parameter(torch.empty(0))
+        self.adaptive_weight = Parameter(torch.empty(0))
+        self.adaptive_bias = Parameter(torch.empty(0)) if adaptive_bias is not None else None
+        self.initial_global_weight_atten = Parameter(torch.empty(0), requires_grad=False)
+        self.initial_adaptive_weight = Parameter(torch.empty(0), requires_grad=False)
+        # the constructor always receives a RAW module weight: add the stack
+        # dim unconditionally (re-inits via init_training_weights pass
+        # already-stacked tensors)
+        self.init_training_weights(global_weight.unsqueeze(-1),
+                                   adaptive_bias=adaptive_bias)
+
+    _assign = staticmethod(AdaptiveBase._assign)
+
+    @torch.no_grad()
+    def init_training_weights(self, global_weight=None, global_weight_atten=None,
+                              adaptive_weight=None, adaptive_bias=None) -> None:
+        if global_weight is None:
+            global_weight = self.global_weight.data
+        self._assign(self.global_weight, global_weight.detach())
+        self.global_weight.requires_grad = False
+
+        k = self.global_weight.shape[-1]
+        if global_weight_atten is None:
+            global_weight_atten = torch.ones(k, device=self.global_weight.device) \
+                                  * self.atten_default
+        self._assign(self.global_weight_atten, global_weight_atten.detach())
+        self._assign(self.initial_global_weight_atten, global_weight_atten.detach())
+        self.global_weight_atten.requires_grad = True
+
+        if self.adaptive_weight.numel() == 0:   # created once, then persists
+            if adaptive_weight is None:
+                adaptive_weight = ((1.0 - self.global_weight_atten.data)
+                                   * self.global_weight.data).sum(dim=-1)
+            self._assign(self.adaptive_weight, adaptive_weight.detach())
+        elif adaptive_weight is not None:
+            self._assign(self.adaptive_weight, adaptive_weight.detach())
+        self.adaptive_weight.requires_grad = True
+        self._assign(self.initial_adaptive_weight, self.adaptive_weight.detach())
+
+        if self.adaptive_bias is not None and adaptive_bias is not None:
+            self._assign(self.adaptive_bias, adaptive_bias.detach())
+            self.adaptive_bias.requires_grad = True
+
+    def composed_weight(self) -> torch.Tensor:
+        return (self.global_weight_atten * self.global_weight).sum(dim=-1) \
+               + self.adaptive_weight
+
+    def drift_pairs(self):
+        return [(self.global_weight_atten, self.initial_global_weight_atten),
+                (self.adaptive_weight, self.initial_adaptive_weight)]
+
+
+class StackedAttenLinear(StackedAttenBase):
+    def forward(self, data: torch.Tensor) -> torch.Tensor:
+        return F.linear(data, self.composed_weight(), self.adaptive_bias)
+
+
+class StackedAttenConv2d(StackedAttenBase):
+    def __init__(self, global_weight, stride=1, padding=0, **kwargs):
+        super().__init__(global_weight, **kwargs)
+        self.stride = stride
+        self.padding = padding
+
+    def forward(self, data: torch.Tensor) -> torch.Tensor:
+        return F.conv2d(data, self.composed_weight(), self.adaptive_bias,
+                        stride=self.stride, padding=self.padding)
+
+
+ADAPTIVE_TYPES = ADAPTIVE_TYPES + (StackedAttenLinear, StackedAttenConv2d)
+
+
+def convert_to_stacked(net: nn.Module, atten_default: float = 0.0) -> int:
+    count = 0
+    for name, module in list(net.named_modules()):
+        if isinstance(module, (nn.Linear, nn.Conv2d)):
+            if not all(p.requires_grad for p in module.parameters()):
+                continue
+            if isinstance(module, nn.Linear):
+                new = StackedAttenLinear(global_weight=module.weight,
+                                         adaptive_bias=module.bias,
+                                         atten_default=atten_default)
+            else:
+                new = StackedAttenConv2d(global_weight=module.weight,
+                                         adaptive_bias=module.bias,
+                                         atten_default=atten_default,
+                                         stride=module.stride,
+                                         padding=module.padding)
+            parent = net
+            parts = name.split(".")
+            for p in parts[:-1]:
+                parent = getattr(parent, p)
+            setattr(parent, parts[-1], new)
+            count += 1
+    return count
