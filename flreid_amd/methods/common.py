@@ -54,6 +54,13 @@ class BaseReIDOperator(OperatorModule):
         if capturable and "capturable" in defaults:
             defaults["capturable"] = True
         params = [p for p in model.net.parameters() if p.requires_grad]
+        # a dispatch may have re-shaped a parameter (stacked-atten variant):
+        # its optimizer state of the old shape must be dropped
+        for p in params:
+            state = self.optimizer.state.get(p)
+            if state and any(torch.is_tensor(v) and v.dim() > 0
+                             and v.shape != p.shape for v in state.values()):
+                del self.optimizer.state[p]
         group = {"params": params, **defaults}
         if capturable:
             group["capturable"] = True

@@ -237,8 +237,10 @@ class Model(ModelModule):
                         # rebind the parameter's storage
                         mod = self.net.get_submodule(n.rsplit(".", 1)[0])
                         leaf = n.rsplit(".", 1)[1]
-                        getattr(mod, leaf).data = p.detach().clone().to(
-                            dst.device, dst.dtype)
+                        target = getattr(mod, leaf)
+                        target.data = p.detach().clone().to(dst.device, dst.dtype)
+                        if hasattr(target, "grad"):
+                            target.grad = None
 
     def composed_upload(self) -> Dict[str, torch.Tensor]:
         """{name.global_weight: atten⊙W_glob + W_adapt} — what the client

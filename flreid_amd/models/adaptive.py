@@ -56,11 +56,13 @@ class AdaptiveBase(nn.Module):
     @staticmethod
     def _assign(param: Parameter, value: torch.Tensor) -> None:
         """In-place when shapes match (keeps optimizer state and captured
-        hipGraph pointers valid across per-round re-inits); rebind otherwise."""
+        hipGraph pointers valid across per-round re-inits); rebind otherwise
+        (clearing any stale gradient of the old shape)."""
         if param.data.shape == value.shape:
             param.data.copy_(value)
         else:
             param.data = value.detach().clone().to(param.device)
+            param.grad = None
 
     @torch.no_grad()
     def init_training_weights(self, global_weight=None, global_weight_atten=None,
