@@ -262,3 +262,9 @@ def cmc_map(query_features, query_labels, gallery_features, gallery_labels,
                                      query_camera_labels, gallery_camera_labels)
     return ref.cmc_map(query_features, query_labels, gallery_features,
                        gallery_labels, query_camera_labels, gallery_camera_labels)
+
+
+def window_attention(q, k, v, bias, mask, scale, dropout=None):
+    """Swin window MHSA (K3) — eager for now; the fused MFMA kernel plugs in
+    here."""
+    return ref.window_attention(q, k, v, bias, mask, scale, dropout)

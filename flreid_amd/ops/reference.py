@@ -292,3 +292,28 @@ def cmc_map_from_sims(sims: torch.Tensor, query_labels: torch.Tensor,
     ap = contrib.sum(dim=1) / n_good.clamp(min=1)
     ap = torch.where(has_good, ap, torch.zeros_like(ap))
     return cmc.cpu(), float(ap.sum().item() / q)
+
+
+# ---------------------------------------------------------------------------
+# Swin window attention (ref:models/swin_transformer.py:255-286)
+# ---------------------------------------------------------------------------
+
+def window_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     bias: torch.Tensor, mask: Optional[torch.Tensor],
+                     scale: float, dropout=None) -> torch.Tensor:
+    """Per-window MHSA: softmax(q·kᵀ·scale + rel-pos-bias [+ shift mask])·v.
+
+    q/k/v: [B·nW, H, N, D]; bias: [H, N, N]; mask: [nW, N, N] or None.
+    Eager composition — the fused CDNA4 kernel (K3) takes over on GPU.
+    """
+    bn, h, n, _d = q.shape
+    attn = (q * scale) @ k.transpose(-2, -1) + bias.unsqueeze(0)
+    if mask is not None:
+        nw = mask.shape[0]
+        attn = attn.view(bn // nw, nw, h, n, n) \
+            + mask.unsqueeze(1).unsqueeze(0).to(attn.dtype)
+        attn = attn.view(bn, h, n, n)
+    attn = F.softmax(attn, dim=-1)
+    if dropout is not None:
+        attn = dropout(attn)
+    return attn @ v
