@@ -1,0 +1,47 @@
+"""Offline analysis layer (ref:analyse/)."""
+
+import torch
+
+from flreid_amd.analyse.accuracy import accuracy_on_round, accuracy_per_task
+from flreid_amd.analyse.forgetting import forgetting_per_client, mean_forgetting
+
+RECORDS = {"data": {
+    "client-0": {
+        "10": {"task-0-0": {"val_rank_1": 0.8, "val_map": 0.6}},
+        "20": {"task-0-0": {"val_rank_1": 0.6, "val_map": 0.5},
+               "task-0-1": {"val_rank_1": 1.0, "val_map": 0.9}},
+    },
+    "client-1": {
+        "10": {"task-1-0": {"val_rank_1": 0.4, "val_map": 0.3}},
+        "20": {"task-1-0": {"val_rank_1": 0.2, "val_map": 0.1}},
+    },
+}}
+
+
+def test_accuracy_on_round():
+    curve = accuracy_on_round(RECORDS, "val_rank_1")
+    assert curve[10] == (0.8 + 0.4) / 2
+    assert abs(curve[20] - (0.6 + 1.0 + 0.2) / 3) < 1e-9
+
+
+def test_accuracy_per_task():
+    per = accuracy_per_task(RECORDS, "val_map")
+    assert per["task-0-0"][20] == 0.5
+
+
+def test_forgetting():
+    fpc = forgetting_per_client(RECORDS, "val_rank_1")
+    # client-0 task-0-0: peak .8 final .6 -> .2 (task-0-1 single point skipped)
+    assert abs(fpc["client-0"] - 0.2) < 1e-9
+    assert abs(fpc["client-1"] - 0.2) < 1e-9
+    assert abs(mean_forgetting(RECORDS, "val_rank_1") - 0.2) < 1e-9
+
+
+def test_grad_cam_runs():
+    from flreid_amd.analyse.visualize import grad_cam
+    from flreid_amd.models import nets
+    net = nets["resnet18"](num_classes=8, last_stride=1, neck="bnneck")
+    img = torch.randn(1, 3, 64, 32)
+    cam = grad_cam(net, net.base.layer4[-1], img)
+    assert cam.shape == (64, 32)
+    assert float(cam.min()) >= 0.0 and float(cam.max()) <= 1.0

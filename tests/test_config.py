@@ -44,7 +44,7 @@ def test_reference_configs_parse():
     count = 0
     for dirpath, _dirs, files in os.walk(root):
         for f in files:
-            if f.endswith(".yaml") and f != "common.yaml":
+            if f.endswith(".yaml") and not f.startswith("common"):
                 merged = load_experiments(common, os.path.join(dirpath, f))[0]
                 assert "exp_name" in merged and "exp_method" in merged
                 count += 1
