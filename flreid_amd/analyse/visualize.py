@@ -29,6 +29,9 @@ def grad_cam(model: torch.nn.Module, target_layer: torch.nn.Module,
     try:
         was_training = model.training
         model.train()        # dual-output forward provides the class scores
+        if image.shape[0] == 1:
+            # batch of 2 keeps train-mode BatchNorm happy; CAM reads sample 0
+            image = torch.cat([image, image], dim=0)
         score, _feat = model(image)
         if class_index is None:
             class_index = int(score.argmax(dim=1))
