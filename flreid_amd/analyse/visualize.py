@@ -34,7 +34,7 @@ def grad_cam(model: torch.nn.Module, target_layer: torch.nn.Module,
             image = torch.cat([image, image], dim=0)
         score, _feat = model(image)
         if class_index is None:
-            class_index = int(score.argmax(dim=1))
+            class_index = int(score[0].argmax())
         model.zero_grad(set_to_none=True)
         score[0, class_index].backward()
         a, g = acts["v"], grads["v"]
