@@ -79,7 +79,9 @@ def build_configs(args, world_size, rank):
                        "last_stride": 1, "neck": "bnneck",
                        "atten_default": 0.9, "lambda_l1": 1e-4,
                        "lambda_k": 2000,
-                       "fine_tuning": ["base.layer4", "classifier"]},
+                       "fine_tuning": (["base.layers.3", "classifier"]
+                                       if args.model.startswith("swin")
+                                       else ["base.layer4", "classifier"])},
         "criterion_opts": {"name": "cross_entropy",
                            "num_classes": args.num_classes, "epsilon": 0.1},
         "optimizer_opts": {"name": "adam", "lr": 1e-3, "weight_decay": 1e-5},

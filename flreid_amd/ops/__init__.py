@@ -265,6 +265,26 @@ def cmc_map(query_features, query_labels, gallery_features, gallery_labels,
 
 
 def window_attention(q, k, v, bias, mask, scale, dropout=None):
-    """Swin window MHSA (K3) — eager for now; the fused MFMA kernel plugs in
-    here."""
+    """Swin window MHSA (K3): fused HIP kernel for the no-grad GPU path (the
+    frozen-backbone prototype capture + eval forward — the FedSTIL-Swin hot
+    loop); eager composition where autograd or dropout is needed."""
+    if (q.is_cuda and not torch.is_grad_enabled() and dropout is None
+            and q.shape[-2] <= 64 and q.shape[-1] <= 64):
+        ext = _ext_or_raise("window_attn_fwd")
+        if ext is not None:
+            bw, h, n, d = q.shape
+            qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+            bias_c = bias.detach().contiguous().float()
+            if mask is not None:
+                mask_c = mask.detach().contiguous().float()
+                nw = mask_c.shape[0]
+                mask_ptr = mask_c.data_ptr()
+            else:
+                mask_c, nw, mask_ptr = None, 1, 0
+            out = torch.empty_like(qc)
+            ext.window_attn_fwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                                bias_c.data_ptr(), mask_ptr, out.data_ptr(),
+                                bw, h, n, d, nw, float(scale), _dt(qc),
+                                _stream())
+            return out
     return ref.window_attention(q, k, v, bias, mask, scale, dropout)

@@ -25,6 +25,10 @@ extern "C" void flreid_pairwise(const float*, const float*, const float*,
                                 int64_t, int, hipStream_t);
 extern "C" void flreid_rowsq(const float*, float*, int64_t, int64_t,
                              hipStream_t);
+extern "C" void flreid_window_attn_fwd(const void*, const void*, const void*,
+                                       const float*, const float*, void*,
+                                       int64_t, int, int, int, int, float,
+                                       int, hipStream_t);
 }  // namespace flreid
 
 namespace py = pybind11;
@@ -81,4 +85,14 @@ PYBIND11_MODULE(_flreid_hip, m) {
     flreid::flreid_rowsq((const float*)x, (float*)out, rows, cols,
                          as_stream(stream));
   });
+
+  m.def("window_attn_fwd",
+        [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t bias,
+           uintptr_t mask, uintptr_t out, int64_t BW, int H, int N, int D,
+           int nW, float scale, int dtype, uintptr_t stream) {
+          flreid::flreid_window_attn_fwd(
+              (const void*)q, (const void*)k, (const void*)v,
+              (const float*)bias, (const float*)mask, (void*)out, BW, H, N, D,
+              nW, scale, dtype, as_stream(stream));
+        });
 }

@@ -125,3 +125,36 @@ def test_cmc_map_gpu_matches_cpu():
     cmc_gpu, map_gpu = ops.cmc_map(qf.cuda(), ql, gf.cuda(), gl)
     assert torch.allclose(cmc_cpu, cmc_gpu, atol=1e-9)
     assert abs(map_cpu - map_gpu) < 1e-6
+
+
+def test_window_attention_fused_vs_eager():
+    torch.manual_seed(0)
+    bw, h, n, d, nw = 32, 3, 49, 32, 4
+    q = torch.randn(bw, h, n, d, device="cuda")
+    k = torch.randn(bw, h, n, d, device="cuda")
+    v = torch.randn(bw, h, n, d, device="cuda")
+    bias = torch.randn(h, n, n, device="cuda")
+    mask = torch.zeros(nw, n, n, device="cuda")
+    mask[:, : n // 2, n // 2:] = -100.0
+    with torch.no_grad():
+        out = ops.window_attention(q, k, v, bias, mask, 0.17)
+        expected = ref.window_attention(q, k, v, bias, mask, 0.17)
+    assert torch.allclose(out, expected, atol=1e-4, rtol=1e-4)
+    # no-mask path
+    with torch.no_grad():
+        out2 = ops.window_attention(q, k, v, bias, None, 0.17)
+        exp2 = ref.window_attention(q, k, v, bias, None, 0.17)
+    assert torch.allclose(out2, exp2, atol=1e-4, rtol=1e-4)
+
+
+def test_window_attention_bf16():
+    torch.manual_seed(1)
+    q = torch.randn(16, 4, 49, 32, device="cuda").bfloat16()
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    bias = torch.randn(4, 49, 49, device="cuda")
+    with torch.no_grad():
+        out = ops.window_attention(q, k, v, bias, None, 0.2)
+        expected = ref.window_attention(q.float(), k.float(), v.float(),
+                                        bias, None, 0.2)
+    assert torch.allclose(out.float(), expected, atol=3e-2, rtol=3e-2)
