@@ -160,11 +160,22 @@ def main():
     # max over ranks
     elapsed = ctx.all_reduce_scalar(elapsed, op="max")
 
+    # Rank-1 on the synthetic tasks (the BASELINE metric pairs throughput
+    # with Rank-1) — measured OUTSIDE the timed region
+    rank1_sum = 0.0
+    for client in clients:
+        task = client.task_pipeline.get_task(0)
+        cmc, _mAP, _rep = client.validate(
+            task_name=task["task_name"], query_loader=task["query_loader"],
+            gallery_loader=task["gallery_loaders"], device=stage.device)
+        rank1_sum += float(cmc[0])
+    rank1 = ctx.all_reduce_scalar(rank1_sum) / max(1, n_clients)
+
     if ctx.is_rank0():
         total_images = imgs_per_client * n_clients * args.steps
         value = total_images / elapsed
         result = {
-            "metric": "images/sec (whole node) per fed round, FedSTIL ResNet50",
+            "metric": f"images/sec (whole node) per fed round, {args.method} {args.model}",
             "value": round(value, 2),
             "unit": "images/sec",
             "n_gpus": max(1, ctx.world_size),
@@ -176,6 +187,7 @@ def main():
             "vs_baseline": None,
             "dtype": "bf16" if device_is_cuda else "fp32",
             "data": "synthetic",
+            "rank1": round(rank1, 4),
             "config": {
                 "model": f"{args.model}-{args.method}",
                 "global_batch": args.batch * n_clients,

@@ -307,13 +307,13 @@ def window_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     Eager composition — the fused CDNA4 kernel (K3) takes over on GPU.
     """
     bn, h, n, _d = q.shape
-    attn = (q * scale) @ k.transpose(-2, -1) + bias.unsqueeze(0)
+    attn = (q * scale) @ k.transpose(-2, -1) + bias.unsqueeze(0).to(q.dtype)
     if mask is not None:
         nw = mask.shape[0]
         attn = attn.view(bn // nw, nw, h, n, n) \
             + mask.unsqueeze(1).unsqueeze(0).to(attn.dtype)
         attn = attn.view(bn, h, n, n)
-    attn = F.softmax(attn, dim=-1)
+    attn = F.softmax(attn.float(), dim=-1).to(q.dtype)
     if dropout is not None:
         attn = dropout(attn)
     return attn @ v
