@@ -45,18 +45,41 @@ __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
 
   f32x16 acc = {};
 
-  const int lc = tid & (BK - 1);        // load col 0..31
-  const int lr0 = tid >> 5;             // load row base 0..7 (8 rows/pass)
+  // float4 staging: 8 lanes cover one 32-float K-row; interior blocks skip
+  // every bounds check and the int64 per-element address math (PMC showed
+  // the guarded scalar loads made the kernel VALU-bound: 4.5 VALU/MFMA)
+  const int lc4 = (tid & 7) * 4;        // load col 0,4,...,28
+  const int lr0 = tid >> 3;             // load row 0..31 (32 rows/pass)
+  const bool interior = (row0 + BM <= M) && (col0 + BN <= N) && (D % 4 == 0);
 
   for (int64_t k0 = 0; k0 < D; k0 += BK) {
+    if (interior && k0 + BK <= D) {
+      // vector global loads; scalar LDS stores (the padded row stride keeps
+      // b32 column reads conflict-free but breaks 16B LDS alignment)
+      const float* arow = A + (row0 + lr0) * D + k0 + lc4;
+      const float* brow = B + (col0 + lr0) * D + k0 + lc4;
 #pragma unroll
-    for (int r = 0; r < BM; r += 8) {
-      const int lr = lr0 + r;
-      const int64_t ar = row0 + lr;
-      const int64_t bc = col0 + lr;
-      const int64_t kk = k0 + lc;
-      ldsA[lr][lc] = (ar < M && kk < D) ? A[ar * D + kk] : 0.0f;
-      ldsB[lr][lc] = (bc < N && kk < D) ? B[bc * D + kk] : 0.0f;
+      for (int r = 0; r < BM; r += 32) {
+        const float4 va = *(const float4*)(arow + (int64_t)r * D);
+        const float4 vb = *(const float4*)(brow + (int64_t)r * D);
+        ldsA[lr0 + r][lc4 + 0] = va.x; ldsA[lr0 + r][lc4 + 1] = va.y;
+        ldsA[lr0 + r][lc4 + 2] = va.z; ldsA[lr0 + r][lc4 + 3] = va.w;
+        ldsB[lr0 + r][lc4 + 0] = vb.x; ldsB[lr0 + r][lc4 + 1] = vb.y;
+        ldsB[lr0 + r][lc4 + 2] = vb.z; ldsB[lr0 + r][lc4 + 3] = vb.w;
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < BM; r += 32) {
+        const int lr = lr0 + r;
+        const int64_t ar = row0 + lr;
+        const int64_t bc = col0 + lr;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const int64_t kk = k0 + lc4 + c;
+          ldsA[lr][lc4 + c] = (ar < M && kk < D) ? A[ar * D + kk] : 0.0f;
+          ldsB[lr][lc4 + c] = (bc < N && kk < D) ? B[bc * D + kk] : 0.0f;
+        }
+      }
     }
     __syncthreads();
 
