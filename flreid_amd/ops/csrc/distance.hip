@@ -22,11 +22,15 @@ namespace flreid {
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
 
-constexpr int BM = 64;      // block tile rows
-constexpr int BN = 64;      // block tile cols
+constexpr int BM = 128;     // block tile rows
+constexpr int BN = 128;     // block tile cols
 constexpr int BK = 32;      // K panel
 constexpr int PAD = 1;      // LDS row padding (33 floats/row)
 
+// 128×128 tile, 4 waves, each wave owns a 64×64 quadrant = 2×2 MFMA 32×32
+// sub-tiles (4 independent accumulators pipeline the 64-cycle f32 MFMA).
+// Arithmetic intensity 32 flops/byte — the 64×64 version measured HBM-bound
+// at 117 TF; this tile halves the HBM traffic.
 template <int MODE>
 __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
@@ -40,22 +44,23 @@ __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;            // 4 waves
-  const int wr = (wave >> 1) * 32;      // wave sub-tile row offset (0/32)
-  const int wc = (wave & 1) * 32;       // wave sub-tile col offset (0/32)
+  const int wr = (wave >> 1) * 64;      // wave quadrant row offset (0/64)
+  const int wc = (wave & 1) * 64;       // wave quadrant col offset (0/64)
 
-  f32x16 acc = {};
+  f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
 
   // float4 staging: 8 lanes cover one 32-float K-row; interior blocks skip
   // every bounds check and the int64 per-element address math (PMC showed
-  // the guarded scalar loads made the kernel VALU-bound: 4.5 VALU/MFMA)
+  // guarded scalar loads made the kernel VALU-bound: 4.5 VALU/MFMA)
   const int lc4 = (tid & 7) * 4;        // load col 0,4,...,28
   const int lr0 = tid >> 3;             // load row 0..31 (32 rows/pass)
   const bool interior = (row0 + BM <= M) && (col0 + BN <= N) && (D % 4 == 0);
 
+  const int fr = lane & 31;             // fragment row/col within 32-tile
+  const int fk = lane >> 5;             // fragment k (0/1)
+
   for (int64_t k0 = 0; k0 < D; k0 += BK) {
     if (interior && k0 + BK <= D) {
-      // vector global loads; scalar LDS stores (the padded row stride keeps
-      // b32 column reads conflict-free but breaks 16B LDS alignment)
       const float* arow = A + (row0 + lr0) * D + k0 + lc4;
       const float* brow = B + (col0 + lr0) * D + k0 + lc4;
 #pragma unroll
@@ -85,23 +90,37 @@ __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
 
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
-      const float a = ldsA[wr + (lane & 31)][kk + (lane >> 5)];
-      const float b = ldsB[wc + (lane & 31)][kk + (lane >> 5)];
-      acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+      const float a0 = ldsA[wr + fr][kk + fk];
+      const float a1 = ldsA[wr + 32 + fr][kk + fk];
+      const float b0 = ldsB[wc + fr][kk + fk];
+      const float b1 = ldsB[wc + 32 + fr][kk + fk];
+      acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+      acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
     }
     __syncthreads();
   }
 
   // C/D mapping (guide §3): col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
-  const int jc = col0 + wc + (lane & 31);
 #pragma unroll
-  for (int reg = 0; reg < 16; ++reg) {
-    const int ir = row0 + wr + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
-    if (ir < M && jc < N) {
-      float v = acc[reg];
-      if (MODE == 1) v = 1.0f - v;
-      if (MODE == 2) v = aa[ir] + bb[jc] - 2.0f * v;
-      OUT[(int64_t)ir * N + jc] = v;
+  for (int ti = 0; ti < 2; ++ti) {
+#pragma unroll
+    for (int tj = 0; tj < 2; ++tj) {
+      const f32x16 acc = ti == 0 ? (tj == 0 ? acc00 : acc01)
+                                 : (tj == 0 ? acc10 : acc11);
+      const int64_t jc = col0 + wc + tj * 32 + (lane & 31);
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int64_t ir = row0 + wr + ti * 32 + (reg & 3) + 8 * (reg >> 2)
+                           + 4 * (lane >> 5);
+        if (ir < M && jc < N) {
+          float v = acc[reg];
+          if (MODE == 1) v = 1.0f - v;
+          if (MODE == 2) v = aa[ir] + bb[jc] - 2.0f * v;
+          OUT[ir * N + jc] = v;
+        }
+      }
     }
   }
 }
