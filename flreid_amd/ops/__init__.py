@@ -246,9 +246,48 @@ l1_drift = ref.l1_drift
 l1_drift_fused = ref.l1_drift_fused
 
 
+class _TripletHardFn(torch.autograd.Function):
+    """Fused batch-hard euclidean triplet (margin mode) — K5."""
+
+    @staticmethod
+    def forward(ctx, feature, target, margin, ext):
+        f = feature.detach().contiguous().float()
+        n, d = f.shape
+        norms = torch.empty(n, device=f.device, dtype=torch.float32)
+        ext.rowsq(f.data_ptr(), norms.data_ptr(), n, d, _stream())
+        row_loss = torch.empty(n, device=f.device, dtype=torch.float32)
+        p_idx = torch.empty(n, device=f.device, dtype=torch.int32)
+        n_idx = torch.empty(n, device=f.device, dtype=torch.int32)
+        ext.triplet_fwd(f.data_ptr(), norms.data_ptr(),
+                        target.contiguous().data_ptr(), row_loss.data_ptr(),
+                        p_idx.data_ptr(), n_idx.data_ptr(), n, d,
+                        float(margin), _stream())
+        ctx.save_for_backward(f, row_loss, p_idx, n_idx)
+        ctx.ext = ext
+        return row_loss.mean()
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        f, row_loss, p_idx, n_idx = ctx.saved_tensors
+        n, d = f.shape
+        grad = torch.zeros_like(f)
+        # coeff = go * (1/N) * 2  (hinge mean + d(dist)/d(f) factor)
+        coeff = float(grad_out) * 2.0 / n
+        ctx.ext.triplet_bwd(f.data_ptr(), row_loss.data_ptr(),
+                            p_idx.data_ptr(), n_idx.data_ptr(),
+                            grad.data_ptr(), n, d, coeff, _stream())
+        return grad, None, None, None
+
+
 def triplet_loss(feature: torch.Tensor, target: torch.Tensor,
                  margin: Optional[float] = 0.3, norm_feat: bool = False,
                  hard_mining: bool = True) -> torch.Tensor:
+    if (feature.is_cuda and hard_mining and not norm_feat
+            and margin is not None and margin > 0 and feature.dim() == 2):
+        ext = _ext_or_raise("triplet_fwd")
+        if ext is not None:
+            return _TripletHardFn.apply(feature.float(), target.long(),
+                                        margin, ext)
     return ref.triplet_loss(feature, target, margin, norm_feat, hard_mining)
 
 

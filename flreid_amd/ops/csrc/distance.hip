@@ -39,17 +39,10 @@ __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
   __shared__ float ldsA[BM][BK + PAD];
   __shared__ float ldsB[BN][BK + PAD];
 
-  // XCD-aware bijective remap (guide T1): linearised block id -> contiguous
-  // chunks per XCD so neighbouring tiles (shared A/B panels) hit one L2
-  const int nbx = gridDim.x, nby = gridDim.y;
-  int bid = blockIdx.y * nbx + blockIdx.x;
-  {
-    const int nwg = nbx * nby, q = nwg / 8, r8 = nwg % 8;
-    const int xcd = bid % 8, pos = bid / 8;
-    bid = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q) + pos;
-  }
-  const int64_t row0 = (int64_t)(bid % nbx) * BM;
-  const int64_t col0 = (int64_t)(bid / nbx) * BN;
+  // note: an XCD-aware block swizzle (guide T1) measured neutral here —
+  // x-major dispatch already gives consecutive blocks a shared B panel
+  const int64_t row0 = (int64_t)blockIdx.x * BM;
+  const int64_t col0 = (int64_t)blockIdx.y * BN;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;            // 4 waves

@@ -29,6 +29,12 @@ extern "C" void flreid_window_attn_fwd(const void*, const void*, const void*,
                                        const float*, const float*, void*,
                                        int64_t, int, int, int, int, float,
                                        int, hipStream_t);
+extern "C" void flreid_triplet_fwd(const float*, const float*, const int64_t*,
+                                   float*, int*, int*, int, int, float,
+                                   hipStream_t);
+extern "C" void flreid_triplet_bwd(const float*, const float*, const int*,
+                                   const int*, float*, int, int, float,
+                                   hipStream_t);
 }  // namespace flreid
 
 namespace py = pybind11;
@@ -94,5 +100,24 @@ PYBIND11_MODULE(_flreid_hip, m) {
               (const void*)q, (const void*)k, (const void*)v,
               (const float*)bias, (const float*)mask, (void*)out, BW, H, N, D,
               nW, scale, dtype, as_stream(stream));
+        });
+
+  m.def("triplet_fwd",
+        [](uintptr_t F, uintptr_t norms, uintptr_t labels, uintptr_t row_loss,
+           uintptr_t p_idx, uintptr_t n_idx, int N, int D, float margin,
+           uintptr_t stream) {
+          flreid::flreid_triplet_fwd((const float*)F, (const float*)norms,
+                                     (const int64_t*)labels, (float*)row_loss,
+                                     (int*)p_idx, (int*)n_idx, N, D, margin,
+                                     as_stream(stream));
+        });
+
+  m.def("triplet_bwd",
+        [](uintptr_t F, uintptr_t row_loss, uintptr_t p_idx, uintptr_t n_idx,
+           uintptr_t grad, int N, int D, float coeff, uintptr_t stream) {
+          flreid::flreid_triplet_bwd((const float*)F, (const float*)row_loss,
+                                     (const int*)p_idx, (const int*)n_idx,
+                                     (float*)grad, N, D, coeff,
+                                     as_stream(stream));
         });
 }

@@ -158,3 +158,25 @@ def test_window_attention_bf16():
         expected = ref.window_attention(q.float(), k.float(), v.float(),
                                         bias, None, 0.2)
     assert torch.allclose(out.float(), expected, atol=3e-2, rtol=3e-2)
+
+
+def test_triplet_fused_fwd_bwd_matches_eager():
+    torch.manual_seed(0)
+    feat = torch.randn(64, 512, device="cuda", requires_grad=True)
+    target = torch.randint(0, 16, (64,), device="cuda")
+    loss = ops.triplet_loss(feat, target, margin=0.3, hard_mining=True)
+    ref_feat = feat.detach().clone().requires_grad_(True)
+    ref_loss = ref.triplet_loss(ref_feat, target, margin=0.3, hard_mining=True)
+    assert torch.allclose(loss, ref_loss, atol=1e-4, rtol=1e-4)
+    (loss * 1.7).backward()
+    (ref_loss * 1.7).backward()
+    assert torch.allclose(feat.grad, ref_feat.grad, atol=1e-4, rtol=1e-4)
+
+
+def test_triplet_fused_bf16_autocast_path():
+    feat32 = torch.randn(32, 256, device="cuda", requires_grad=True)
+    target = torch.randint(0, 8, (32,), device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        loss = ops.triplet_loss(feat32, target, margin=0.3, hard_mining=True)
+    loss.backward()
+    assert torch.isfinite(feat32.grad).all()
