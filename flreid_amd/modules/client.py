@@ -52,6 +52,8 @@ class ClientModule:
             if default_value is not None:
                 return default_value
             raise ValueError("ckpt disabled and no default value")
+        from flreid_amd.runtime.io import before_ckpt_read
+        before_ckpt_read()
         state_path = os.path.join(self.ckpt_path, f"{state_name}.ckpt")
         os.makedirs(self.ckpt_path, exist_ok=True)
         if os.path.exists(state_path):
@@ -63,6 +65,8 @@ class ClientModule:
     def state_exists(self, state_name: str) -> bool:
         if self._ckpt_disabled():
             return False
+        from flreid_amd.runtime.io import before_ckpt_read
+        before_ckpt_read()
         return os.path.exists(os.path.join(self.ckpt_path, f"{state_name}.ckpt"))
 
     def save_state(self, state_name: str, state: Any, cover: bool = False) -> None:
@@ -72,7 +76,8 @@ class ClientModule:
         os.makedirs(self.ckpt_path, exist_ok=True)
         if not cover and os.path.exists(state_path):
             raise ValueError(f"State checkpoint already exists in '{state_path}'.")
-        torch.save(state, state_path)
+        from flreid_amd.runtime.io import save_ckpt
+        save_ckpt(state_path, state)
 
     # ----------------------------------------------------------------- model
     def load_model(self, model_name: str) -> None:

@@ -45,6 +45,8 @@ class ServerModule:
             if default_value is not None:
                 return default_value
             raise ValueError("ckpt disabled and no default value")
+        from flreid_amd.runtime.io import before_ckpt_read
+        before_ckpt_read()
         state_path = os.path.join(self.ckpt_path, f"{state_name}.ckpt")
         os.makedirs(self.ckpt_path, exist_ok=True)
         if os.path.exists(state_path):
@@ -60,7 +62,8 @@ class ServerModule:
         os.makedirs(self.ckpt_path, exist_ok=True)
         if not cover and os.path.exists(state_path):
             raise ValueError(f"State checkpoint already exists in '{state_path}'.")
-        torch.save(state, state_path)
+        from flreid_amd.runtime.io import save_ckpt
+        save_ckpt(state_path, state)
 
     # ----------------------------------------------------------------- model
     def load_model(self, model_name: str) -> None:

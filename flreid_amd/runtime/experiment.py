@@ -36,7 +36,7 @@ from flreid_amd.parallel.comm import FedContext, get_context
 from flreid_amd.runtime.builder import parser_clients, parser_server
 from flreid_amd.runtime.log import ExperimentLog
 from flreid_amd.tools.logger import Logger
-from flreid_amd.tools.utils import clear_cache, same_seeds
+from flreid_amd.tools.utils import clear_cache, params_state_size, same_seeds
 
 
 class ExperimentStage:
@@ -130,6 +130,9 @@ class ExperimentStage:
                           log: ExperimentLog) -> None:
         online = self._sample_online(exp_config, curr_round, client_names)
         val_interval = int(exp_config["exp_opts"]["val_interval"])
+        comm_down = comm_up = 0     # per-round communication bytes
+        # (the reference shipped this accounting as dead code,
+        # ref:tools/utils.py:39-48 — live here)
         persist_comm = bool(exp_config.get("exp_opts", {}).get("persist_comm_ckpts", True))
 
         # ---- dispatch (server -> clients); replicated server state ---------
@@ -157,6 +160,8 @@ class ExperimentStage:
             if persist_comm:
                 server.save_state(f"{curr_round}-{server.server_name}-{cname}",
                                   dispatch_state, True)
+            if dispatch_state is not None:
+                comm_down += params_state_size(dispatch_state)
             del dispatch_state
 
         # ---- local training of owned online clients ------------------------
@@ -193,6 +198,7 @@ class ExperimentStage:
                 # the tensor codec keeps device tensors device-resident;
                 # its ragged-schema fallback cpu-ifies internally
                 local_uploads[cname] = state
+                comm_up += params_state_size(state)
 
         # ---- sync uploads across ranks (ONE gather per round) --------------
         from flreid_amd.parallel.codec import sync_client_states
@@ -202,6 +208,9 @@ class ExperimentStage:
         for cname in online:  # deterministic application order
             if cname in merged:
                 server.set_client_incremental_state(cname, merged[cname])
+
+        log.record(f"comm.{curr_round}",
+                   {"dispatch_bytes": comm_down, "upload_bytes": comm_up})
 
         # ---- aggregate (replicated, deterministic) -------------------------
         from flreid_amd.runtime.hipgraph import dump_phases, phase, phase_timers_enabled

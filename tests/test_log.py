@@ -29,3 +29,14 @@ def test_flush_writes_json(tmp_path):
     log.record("config", {"exp_name": "t"})
     log.flush()
     assert json.loads(path.read_text())["config"]["exp_name"] == "t"
+
+
+def test_async_ckpt_write_then_read(tmp_path, monkeypatch):
+    """Async writer ordering: a read after a submitted write sees the data."""
+    import torch
+    monkeypatch.setenv("FLREID_ASYNC_CKPT", "1")
+    from flreid_amd.runtime import io as rio
+    p = str(tmp_path / "x.ckpt")
+    rio.save_ckpt(p, {"w": torch.ones(4)})
+    rio.before_ckpt_read()
+    assert torch.load(p, weights_only=False)["w"].sum() == 4
