@@ -292,15 +292,35 @@ def triplet_loss(feature: torch.Tensor, target: torch.Tensor,
 
 
 def cmc_map(query_features, query_labels, gallery_features, gallery_labels,
-            query_camera_labels=None, gallery_camera_labels=None):
+            query_camera_labels=None, gallery_camera_labels=None,
+            query_chunk: int = 2048):
     """On GPU the Q×G similarity runs on the MFMA pairwise kernel; the rank
-    statistics stay in (device-side) torch ops."""
-    if query_features.is_cuda:
+    statistics stay in device-side torch ops.  Queries are processed in
+    chunks so million-image galleries (iCaRL config — K8/288 GB HBM) never
+    materialise more than [chunk, G] similarities at once."""
+    if not query_features.is_cuda:
+        return ref.cmc_map(query_features, query_labels, gallery_features,
+                           gallery_labels, query_camera_labels,
+                           gallery_camera_labels)
+    q = query_features.shape[0]
+    g = gallery_features.shape[0]
+    if q <= query_chunk:
         sims = similarity_matrix(query_features, gallery_features)
         return ref.cmc_map_from_sims(sims, query_labels, gallery_labels,
                                      query_camera_labels, gallery_camera_labels)
-    return ref.cmc_map(query_features, query_labels, gallery_features,
-                       gallery_labels, query_camera_labels, gallery_camera_labels)
+    total_cmc = torch.zeros(g, dtype=torch.float64)
+    total_ap = 0.0
+    for s0 in range(0, q, query_chunk):
+        s1 = min(q, s0 + query_chunk)
+        sims = similarity_matrix(query_features[s0:s1], gallery_features)
+        cmc, mAP = ref.cmc_map_from_sims(
+            sims, query_labels[s0:s1], gallery_labels,
+            query_camera_labels[s0:s1] if query_camera_labels is not None else None,
+            gallery_camera_labels)
+        n = s1 - s0
+        total_cmc += cmc * n
+        total_ap += mAP * n
+    return total_cmc / q, total_ap / q
 
 
 def window_attention(q, k, v, bias, mask, scale, dropout=None):

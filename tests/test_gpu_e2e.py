@@ -43,15 +43,22 @@ def _exp(method):
             {"client_name": "client-1", "tasks": ["task-1-0", "task-1-1"]},
         ],
     }
-    if method == "fedstil":
-        cfg["model_opts"].update({"atten_default": 0.9, "lambda_l1": 1e-4,
-                                  "lambda_k": 64})
+    if method in ("fedstil", "fedstil-atten"):
+        cfg["model_opts"].update({"atten_default": 0.9 if method == "fedstil" else 0.0,
+                                  "lambda_l1": 1e-4, "lambda_k": 64})
         cfg["server"].update({"distance_calculate_step": 10,
                               "distance_calculate_decay": 0.8})
+    if method in ("ewc", "fedcurv"):
+        cfg["model_opts"]["lambda_penalty"] = 1.0
+    if method == "fedweit":
+        cfg["model_opts"].update({"lambda_l1": 5e-6, "lambda_l2": 1e-3,
+                                  "lambda_mask": 0.0, "kb_cnt": 2})
+    if method == "icarl":
+        cfg["model_opts"].update({"num_classes": 10, "k": 32, "n_classes": 10})
     return cfg
 
 
-@pytest.mark.parametrize("method", ["fedavg", "fedstil"])
+@pytest.mark.parametrize("method", ["fedavg", "fedstil", "fedprox", "fedcurv", "fedweit", "ewc", "icarl", "fedstil-atten"])
 def test_gpu_round(method, tmp_path, monkeypatch):
     from flreid_amd import ops
     assert ops.extension_available()
@@ -68,6 +75,7 @@ def test_gpu_round(method, tmp_path, monkeypatch):
     assert "client-0" in data and "client-1" in data
     r2 = data["client-0"].get("2", {})
     assert any("val_map" in v for v in r2.values())
+    assert "comm" in log.records  # per-round comm-bytes accounting
 
 
 def test_smoke_entry():

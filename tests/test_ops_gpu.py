@@ -180,3 +180,30 @@ def test_triplet_fused_bf16_autocast_path():
         loss = ops.triplet_loss(feat32, target, margin=0.3, hard_mining=True)
     loss.backward()
     assert torch.isfinite(feat32.grad).all()
+
+
+def test_cmc_map_chunked_matches_single():
+    torch.manual_seed(2)
+    qf = ref.l2_normalize(torch.randn(300, 128, device="cuda"))
+    gf = ref.l2_normalize(torch.randn(700, 128, device="cuda"))
+    ql = torch.randint(0, 40, (300,))
+    gl = torch.randint(0, 40, (700,))
+    cmc1, map1 = ops.cmc_map(qf, ql, gf, gl)
+    cmc2, map2 = ops.cmc_map(qf, ql, gf, gl, query_chunk=64)
+    assert torch.allclose(cmc1, cmc2, atol=1e-12)
+    assert abs(map1 - map2) < 1e-9
+
+
+def test_cmc_map_large_gallery():
+    """iCaRL-scale eval (K8): 200k gallery through the MFMA GEMM + device
+    ranking, chunked queries."""
+    torch.manual_seed(3)
+    g, q, d = 200_000, 256, 512
+    gf = ref.l2_normalize(torch.randn(g, d, device="cuda"))
+    qf = ref.l2_normalize(torch.randn(q, d, device="cuda"))
+    gl = torch.randint(0, 5000, (g,))
+    ql = gl[torch.randperm(g)[:q]]        # every query has matches
+    cmc, mAP = ops.cmc_map(qf, ql, gf, gl, query_chunk=128)
+    assert cmc.shape[0] == g
+    assert 0.0 <= mAP <= 1.0
+    assert float(cmc[-1]) <= 1.0 and float(cmc[0]) >= 0.0
