@@ -347,3 +347,30 @@ def window_attention(q, k, v, bias, mask, scale, dropout=None):
                                 _stream())
             return out
     return ref.window_attention(q, k, v, bias, mask, scale, dropout)
+
+
+def adaptive_linear_fwd(x: torch.Tensor, gw: torch.Tensor,
+                        atten: Optional[torch.Tensor],
+                        aw: Optional[torch.Tensor],
+                        bias: Optional[torch.Tensor],
+                        split_layout: int = 1) -> torch.Tensor:
+    """Fused y = x · (atten⊙gw + aw)ᵀ + bias (K2) — bf16 MFMA, no-grad path.
+
+    x bf16 [M, K]; gw/aw fp32 [N, K]; atten fp32 [K]; bias fp32 [N].
+    """
+    ext = _ext_or_raise("adaptive_linear_fwd")
+    if ext is None:
+        theta = ref.adaptive_compose(gw, atten, aw) if atten is not None else gw
+        return (torch.nn.functional.linear(x.float(), theta,
+                                           bias).to(x.dtype))
+    m, k = x.shape
+    n = gw.shape[0]
+    x_c = x.contiguous()
+    out = torch.empty(m, n, device=x.device, dtype=torch.bfloat16)
+    ext.adaptive_linear_fwd(
+        x_c.data_ptr(), gw.contiguous().data_ptr(),
+        aw.contiguous().data_ptr() if aw is not None else 0,
+        atten.contiguous().float().data_ptr() if atten is not None else 0,
+        bias.contiguous().float().data_ptr() if bias is not None else 0,
+        out.data_ptr(), m, n, k, split_layout, _stream())
+    return out

@@ -35,6 +35,10 @@ extern "C" void flreid_triplet_fwd(const float*, const float*, const int64_t*,
 extern "C" void flreid_triplet_bwd(const float*, const float*, const int*,
                                    const int*, float*, int, int, float,
                                    hipStream_t);
+extern "C" void flreid_adaptive_linear_fwd(const void*, const float*,
+                                           const float*, const float*,
+                                           const float*, void*, int, int,
+                                           int, int, hipStream_t);
 }  // namespace flreid
 
 namespace py = pybind11;
@@ -110,6 +114,16 @@ PYBIND11_MODULE(_flreid_hip, m) {
                                      (const int64_t*)labels, (float*)row_loss,
                                      (int*)p_idx, (int*)n_idx, N, D, margin,
                                      as_stream(stream));
+        });
+
+  m.def("adaptive_linear_fwd",
+        [](uintptr_t X, uintptr_t GW, uintptr_t AW, uintptr_t ATTEN,
+           uintptr_t BIAS, uintptr_t OUT, int M, int N, int K,
+           int split_layout, uintptr_t stream) {
+          flreid::flreid_adaptive_linear_fwd(
+              (const void*)X, (const float*)GW, (const float*)AW,
+              (const float*)ATTEN, (const float*)BIAS, (void*)OUT, M, N, K,
+              split_layout, as_stream(stream));
         });
 
   m.def("triplet_bwd",
