@@ -102,6 +102,13 @@ class AdaptiveBase(nn.Module):
 
 class AdaptiveLinear(AdaptiveBase):
     def forward(self, data: torch.Tensor) -> torch.Tensor:
+        if data.dim() == 2:
+            # fused MFMA GEMM with compose-in-prologue (K2); falls back to
+            # compose + F.linear off-GPU or when atten trains
+            return ops.adaptive_linear(data, self.global_weight,
+                                       self.global_weight_atten,
+                                       self.adaptive_weight,
+                                       self.adaptive_bias)
         return F.linear(data, self.composed_weight(), self.adaptive_bias)
 
 

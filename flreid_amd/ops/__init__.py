@@ -374,3 +374,41 @@ def adaptive_linear_fwd(x: torch.Tensor, gw: torch.Tensor,
         bias.contiguous().float().data_ptr() if bias is not None else 0,
         out.data_ptr(), m, n, k, split_layout, _stream())
     return out
+
+
+class _AdaptiveLinearFn(torch.autograd.Function):
+    """Autograd wrapper for the fused K2 GEMM: fwd composes θ in the weight
+    fetch; bwd materialises θ once (compose kernel) for grad_x and computes
+    grad_aw = gᵀ·x.  atten/gw are frozen in FedSTIL (no grads)."""
+
+    @staticmethod
+    def forward(ctx, x, gw, atten, aw, bias):
+        x_bf = x.detach().to(torch.bfloat16).contiguous()
+        out = adaptive_linear_fwd(x_bf, gw.detach(), atten, aw.detach(),
+                                  bias.detach() if bias is not None else None)
+        ctx.save_for_backward(x_bf, gw, atten, aw)
+        ctx.has_bias = bias is not None
+        ctx.x_dtype = x.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x_bf, gw, atten, aw = ctx.saved_tensors
+        g = grad_out.to(torch.bfloat16)
+        theta = adaptive_compose(gw, atten, aw).to(torch.bfloat16)
+        grad_x = (g @ theta).to(ctx.x_dtype)
+        grad_aw = (g.t() @ x_bf).to(aw.dtype)
+        grad_bias = grad_out.sum(0).to(torch.float32) if ctx.has_bias else None
+        return grad_x, None, None, grad_aw, grad_bias
+
+
+def adaptive_linear(x: torch.Tensor, gw: torch.Tensor, atten: torch.Tensor,
+                    aw: torch.Tensor, bias: Optional[torch.Tensor]):
+    """Adaptive linear layer forward: fused MFMA path on GPU (frozen
+    gw/atten), eager compose+linear elsewhere."""
+    if (x.is_cuda and x.dim() == 2 and not gw.requires_grad
+            and not atten.requires_grad and gw.shape[-1] % 32 == 0
+            and extension_available()):
+        return _AdaptiveLinearFn.apply(x, gw, atten, aw, bias)
+    theta = adaptive_compose(gw, atten, aw)
+    return torch.nn.functional.linear(x, theta, bias)

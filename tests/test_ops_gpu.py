@@ -243,3 +243,28 @@ def test_adaptive_linear_fused_classifier_shape():
     out = ops.adaptive_linear_fwd(x, gw, atten, aw, None)
     expected = x.float() @ (atten * gw + aw).t()
     assert torch.allclose(out.float(), expected, atol=0.5, rtol=5e-2)
+
+
+def test_adaptive_linear_autograd_matches_eager():
+    torch.manual_seed(0)
+    m, n, d = 32, 64, 128
+    x = torch.randn(m, d, device="cuda", requires_grad=True)
+    gw = torch.randn(n, d, device="cuda")
+    aw = (torch.randn(n, d, device="cuda") * 0.1).requires_grad_(True)
+    atten = torch.rand(d, device="cuda")
+    bias = torch.zeros(n, device="cuda", requires_grad=True)
+    out = ops.adaptive_linear(x, gw, atten, aw, bias)
+    loss = (out.float() ** 2).mean()
+    loss.backward()
+
+    x2 = x.detach().clone().requires_grad_(True)
+    aw2 = aw.detach().clone().requires_grad_(True)
+    bias2 = bias.detach().clone().requires_grad_(True)
+    theta = (atten * gw + aw2)
+    out2 = torch.nn.functional.linear(x2.bfloat16(), theta.bfloat16(), bias2.bfloat16())
+    loss2 = (out2.float() ** 2).mean()
+    loss2.backward()
+    assert torch.allclose(out.float(), out2.float(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(x.grad, x2.grad, atol=5e-3, rtol=5e-2)
+    assert torch.allclose(aw.grad, aw2.grad, atol=5e-3, rtol=5e-2)
+    assert torch.allclose(bias.grad, bias2.grad, atol=5e-3, rtol=5e-2)
