@@ -16,6 +16,8 @@ from __future__ import annotations
 
 from typing import Any, Dict
 
+import torch
+
 from flreid_amd.methods.common import (
     BaseReIDClient,
     BaseReIDOperator,
@@ -59,6 +61,43 @@ class Client(BaseReIDClient):
 
 
 class Server(ServerModule):
+    #: key holding the uploaded trainable-parameter dict
+    _params_key = "incremental_model_params"
+
+    def collective_aggregate(self, ctx, local_uploads) -> bool:
+        """C1 fast path: each rank pre-scales its OWN
+        clients' cached uploads by k_c and the weighted sum rides ONE
+        bucketed RCCL all-reduce — no state replication.  Stale-upload
+        semantics hold because each rank keeps its own clients' latest
+        uploads in `self.clients`.  Returns True when it handled the round
+        (the driver then skips the gather + local calculate)."""
+        for cname, state in local_uploads.items():
+            if cname in self.clients:
+                self.clients[cname] = state
+        states = {c: s for c, s in self.clients.items() if s}
+        local_k = float(sum(s["train_cnt"] for s in states.values()))
+        total_k = ctx.all_reduce_scalar(local_k)
+        if total_k == 0:
+            return True
+        local_sum = {}
+        for _c, s in states.items():
+            w = s["train_cnt"] / total_k
+            for n, p in s[self._params_key].items():
+                contrib = p.detach().to(torch.float32) * w
+                local_sum[n] = local_sum.get(n, 0) + contrib
+        if not local_sum:
+            # this rank owns no uploads yet: contribute zeros of the right
+            # schema (the model's trainable set)
+            local_sum = {n: torch.zeros_like(p, dtype=torch.float32)
+                         for n, p in self.model.named_parameters()
+                         if p.requires_grad}
+        merged = ctx.weighted_allreduce(local_sum, 1.0)
+        self._apply_merged(merged)
+        return True
+
+    def _apply_merged(self, merged) -> None:
+        self.update_model(merged)
+
     def calculate(self) -> Any:
         states = {c: s for c, s in self.clients.items() if s}
         if not states:

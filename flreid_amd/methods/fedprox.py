@@ -114,6 +114,9 @@ class Server(FedAvgServer):
     """Identical weighted averaging (ref:methods/fedprox.py:488-501), but
     dispatch/collect uses the net's parameter names (no 'net.' prefix)."""
 
+    def _apply_merged(self, merged) -> None:
+        self.model.update_model({"net_params": merged})
+
     def calculate(self) -> Any:
         states = {c: s for c, s in self.clients.items() if s}
         if not states:

@@ -202,20 +202,25 @@ class ExperimentStage:
 
         # ---- sync uploads across ranks (ONE gather per round) --------------
         from flreid_amd.parallel.codec import sync_client_states
-        from flreid_amd.runtime.hipgraph import phase as _phase
-        with _phase("upload_sync"):
-            merged = sync_client_states(self.ctx, local_uploads)
-        for cname in online:  # deterministic application order
-            if cname in merged:
-                server.set_client_incremental_state(cname, merged[cname])
+        from flreid_amd.runtime.hipgraph import dump_phases, phase, phase_timers_enabled
+        collective = getattr(server, "collective_aggregate", None)
+        if collective is not None and self.ctx.is_distributed:
+            # C1 fast path: pre-scaled all-reduce, no upload replication
+            with phase("aggregate"):
+                handled = collective(self.ctx, local_uploads)
+            if not handled:
+                collective = None
+        if collective is None or not self.ctx.is_distributed:
+            with phase("upload_sync"):
+                merged = sync_client_states(self.ctx, local_uploads)
+            for cname in online:  # deterministic application order
+                if cname in merged:
+                    server.set_client_incremental_state(cname, merged[cname])
+            with phase("aggregate"):
+                server.calculate()
 
         log.record(f"comm.{curr_round}",
                    {"dispatch_bytes": comm_down, "upload_bytes": comm_up})
-
-        # ---- aggregate (replicated, deterministic) -------------------------
-        from flreid_amd.runtime.hipgraph import dump_phases, phase, phase_timers_enabled
-        with phase("aggregate"):
-            server.calculate()
         log.sync(self.ctx)
         if phase_timers_enabled():
             print(f"[phases r{curr_round}] {dump_phases()}", flush=True)
