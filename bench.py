@@ -32,6 +32,7 @@ if REPO_ROOT not in sys.path:
 os.environ.setdefault("FLREID_DISABLE_CKPT", "1")
 os.environ.setdefault("FLREID_FAST_CONV", "1")
 os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+os.environ.setdefault("FLREID_GPU_AUGMENT", "1")
 
 import torch  # noqa: E402
 

@@ -48,6 +48,37 @@ class TensorAugment:
                 return img
         return img
 
+    @torch.no_grad()
+    def apply_batch(self, x: torch.Tensor) -> torch.Tensor:
+        """Vectorised batch equivalent of __call__ for [B, C, H, W]."""
+        dev = x.device
+        b, c, h, w = x.shape
+        x = (x.float() - self.mean.to(dev)) / self.std.to(dev)
+        if self.hflip_p:
+            flip = torch.rand(b, device=dev) < self.hflip_p
+            if flip.any():
+                x = torch.where(flip.view(b, 1, 1, 1),
+                                torch.flip(x, dims=[-1]), x)
+        if self.erase_p:
+            do = torch.rand(b, device=dev) < self.erase_p
+            area = h * w
+            target = torch.empty(b, device=dev).uniform_(*self.erase_scale) * area
+            ratio = torch.empty(b, device=dev).uniform_(*self.erase_ratio)
+            eh = (target * ratio).sqrt().round().long().clamp(1, h - 1)
+            ew = (target / ratio).sqrt().round().long().clamp(1, w - 1)
+            y0 = (torch.rand(b, device=dev) * (h - eh).float()).long()
+            x0 = (torch.rand(b, device=dev) * (w - ew).float()).long()
+            rows = torch.arange(h, device=dev).view(1, h, 1)
+            cols = torch.arange(w, device=dev).view(1, 1, w)
+            mask = ((rows >= y0.view(b, 1, 1)) & (rows < (y0 + eh).view(b, 1, 1))
+                    & (cols >= x0.view(b, 1, 1)) & (cols < (x0 + ew).view(b, 1, 1))
+                    & do.view(b, 1, 1)).unsqueeze(1)
+            x = torch.where(mask, torch.randn_like(x), x)
+        if x.shape[-2:] != self.size:
+            x = F.interpolate(x, size=self.size, mode="bilinear",
+                              align_corners=False)
+        return x
+
     def __call__(self, img: torch.Tensor) -> torch.Tensor:
         if not isinstance(img, torch.Tensor):
             img = torch.as_tensor(img, dtype=torch.float32)
@@ -90,3 +121,40 @@ augmentations = {
     "sharp": augmentation_sharp,
     "drastic": augmentation_drastic,
 }
+
+
+class DeviceAugmentLoader:
+    """Batched on-device augmentation (FLREID_GPU_AUGMENT=1).
+
+    Wraps a raw (transform-free) DataLoader: batches move to the device once
+    and the whole augmentation pipeline (normalize / hflip / random-erase /
+    resize) runs vectorised on the GPU — the per-item CPU path measured
+    ~50 ms per 512-image round on MI355X.  Distributions match the per-item
+    transforms; randomness draws from the torch device RNG.
+    """
+
+    def __init__(self, base_loader, augment: TensorAugment, device):
+        self.base = base_loader
+        self.augment = augment
+        self.device = device
+
+    # DataLoader surface used by the framework
+    @property
+    def dataset(self):
+        return self.base.dataset
+
+    @property
+    def batch_size(self):
+        return self.base.batch_size
+
+    @property
+    def drop_last(self):
+        return self.base.drop_last
+
+    def __len__(self):
+        return len(self.base)
+
+    def __iter__(self):
+        for data, pid, cid in self.base:
+            x = data.to(self.device, non_blocking=True)
+            yield self.augment.apply_batch(x), pid, cid

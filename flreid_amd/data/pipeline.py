@@ -16,9 +16,12 @@ from __future__ import annotations
 import os
 from typing import Dict, List
 
+import os
+
+import torch
 from torch.utils.data import DataLoader
 
-from flreid_amd.data.augment import augmentations
+from flreid_amd.data.augment import DeviceAugmentLoader, augmentations
 from flreid_amd.data.loader import ReIDImageDataset
 from flreid_amd.data.synthetic import SyntheticReIDDataset, parse_synthetic_dir
 
@@ -52,6 +55,11 @@ class ReIDTaskPipeline:
                 kwargs["multiprocessing_context"] = lo["multiprocessing_context"]
         return DataLoader(**kwargs)
 
+    @staticmethod
+    def _gpu_augment() -> bool:
+        return (os.environ.get("FLREID_GPU_AUGMENT", "0") == "1"
+                and torch.cuda.is_available())
+
     def _dataset(self, task: str, split: str, transform):
         if self.synthetic_opts is not None:
             so = self.synthetic_opts
@@ -68,6 +76,21 @@ class ReIDTaskPipeline:
                                             std=ao["norm_std"])
         no_aug = augmentations["none"](size=ao["img_size"], mean=ao["norm_mean"],
                                        std=ao["norm_std"])
+        if self._gpu_augment():
+            dev = "cuda"
+            return {
+                "task_name": task,
+                "tr_epochs": self.task_opts["train_epochs"],
+                "tr_loader": DeviceAugmentLoader(
+                    self._make_loader(self._dataset(task, "train", None), True),
+                    tr_aug, dev),
+                "query_loader": DeviceAugmentLoader(
+                    self._make_loader(self._dataset(task, "query", None), False),
+                    no_aug, dev),
+                "gallery_loaders": DeviceAugmentLoader(
+                    self._make_loader(self._dataset(task, "gallery", None), False),
+                    no_aug, dev),
+            }
         return {
             "task_name": task,
             "tr_epochs": self.task_opts["train_epochs"],

