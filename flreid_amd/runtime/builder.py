@@ -73,7 +73,17 @@ def _build_operator(exp_config: Dict) -> Any:
     return model, operator
 
 
+def _actor_seed(exp_config: Dict, actor: str) -> int:
+    import hashlib
+    h = hashlib.sha256(f"{exp_config.get('random_seed', 0)}|{actor}".encode())
+    return int.from_bytes(h.digest()[:4], "little") % (2 ** 31)
+
+
 def parser_server(exp_config: Dict, common_config: Dict) -> ServerModule:
+    # deterministic per-actor init: a client's weights do not depend on which
+    # rank builds it or on sibling build order (placement-invariant sharding)
+    from flreid_amd.tools.utils import same_seeds
+    same_seeds(_actor_seed(exp_config, "server"))
     model, operator = _build_operator(exp_config)
     kwargs = {n: p for n, p in exp_config["server"].items() if n != "server_name"}
     return methods[exp_config["exp_method"]].Server(
@@ -87,10 +97,12 @@ def parser_server(exp_config: Dict, common_config: Dict) -> ServerModule:
 def parser_clients(exp_config: Dict, common_config: Dict,
                    owned_indices: Optional[List[int]] = None) -> List[ClientModule]:
     """Build client objects; with `owned_indices`, only those (rank sharding)."""
+    from flreid_amd.tools.utils import same_seeds
     clients = []
     for idx, client_config in enumerate(exp_config["clients"]):
         if owned_indices is not None and idx not in owned_indices:
             continue
+        same_seeds(_actor_seed(exp_config, client_config["client_name"]))
         model, operator = _build_operator(exp_config)
         task_pipeline = ReIDTaskPipeline(
             task_list=client_config["tasks"],

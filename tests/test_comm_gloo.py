@@ -211,3 +211,94 @@ def _worker_codec(rank, world, port, tmpdir):
 def test_tensor_codec_gather(tmp_path):
     port = _free_port()
     mp.spawn(_worker_codec, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+
+
+def _fedstil_cfg(tmpdir, tag):
+    common = {
+        "datasets_dir": "synthetic://ids=4,train=3,query=2,gallery=3,hw=32x16,idspace=48",
+        "checkpoints_dir": os.path.join(tmpdir, f"ck-{tag}"),
+        "logs_dir": os.path.join(tmpdir, f"lg-{tag}"),
+        "parallel": 1, "device": ["cpu"], "defaults": {},
+    }
+    exp = {
+        "exp_name": "dist-fedstil", "exp_method": "fedstil", "random_seed": 5,
+        "exp_opts": {"comm_rounds": 3, "val_interval": 0, "online_clients": 3,
+                     "initial_validation": False, "persist_comm_ckpts": False},
+        "model_opts": {"name": "resnet18", "num_classes": 64, "last_stride": 1,
+                       "neck": "bnneck", "atten_default": 0.9,
+                       "lambda_l1": 1e-4, "lambda_k": 16,
+                       "fine_tuning": ["classifier"]},
+        "criterion_opts": {"name": "cross_entropy", "num_classes": 64,
+                           "epsilon": 0.1},
+        "optimizer_opts": {"name": "adam", "lr": 1e-3, "weight_decay": 1e-5},
+        "scheduler_opts": {"name": "step_lr", "step_size": 5},
+        "task_opts": {"sustain_rounds": 2, "train_epochs": 1,
+                      "augment_opts": {"level": "none", "img_size": [32, 16],
+                                       "norm_mean": [0.485, 0.456, 0.406],
+                                       "norm_std": [0.229, 0.224, 0.225]},
+                      "loader_opts": {"batch_size": 8, "num_workers": 0,
+                                      "pin_memory": False,
+                                      "persistent_workers": False,
+                                      "multiprocessing_context": None}},
+        "server": {"server_name": "server", "distance_calculate_step": 10,
+                   "distance_calculate_decay": 0.8},
+        "clients": [{"client_name": f"client-{i}", "tasks": [f"task-{i}-0"]}
+                    for i in range(4)],        # 4 clients on 2 ranks,
+    }                                          # 3 online -> partial participation
+    return common, exp
+
+
+def _run_fedstil_rounds(ctx, common, exp):
+    from flreid_amd.runtime.builder import parser_clients, parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+    from flreid_amd.runtime.log import ExperimentLog
+    from flreid_amd.tools.utils import same_seeds
+
+    stage = ExperimentStage(common, [exp], ctx=ctx)
+    same_seeds(exp["random_seed"])
+    log = ExperimentLog(os.path.join(common["logs_dir"], "log.json"))
+    server = parser_server(exp, common)
+    client_names = [c["client_name"] for c in exp["clients"]]
+    owned = [i for i in range(len(client_names))
+             if ctx.owner_of(i) == ctx.rank]
+    clients = parser_clients(exp, common, owned_indices=owned)
+    by_name = {c.client_name: c for c in clients}
+    for r in (1, 2, 3):
+        stage.process_one_round(r, server, by_name, client_names, exp, log)
+    state = server.model.model_state()["global_weight"]
+    return {n: p.clone() for n, p in state.items()}
+
+
+def _worker_fedstil(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        common, exp = _fedstil_cfg(tmpdir, "dist")
+        state = _run_fedstil_rounds(ctx, common, exp)
+        if rank == 0:
+            with open(os.path.join(tmpdir, "fedstil_state.pkl"), "wb") as f:
+                pickle.dump(state, f)
+    finally:
+        destroy_context()
+
+
+@pytest.mark.timeout(600)
+def test_fedstil_distributed_equals_single(tmp_path):
+    """FedSTIL with 4 clients / 2 ranks / partial participation (3 online):
+    the replicated-server + codec-gather path must reproduce the
+    single-process simulator's global weights exactly."""
+    tmpdir = str(tmp_path)
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    from flreid_amd.parallel.comm import FedContext
+    common, exp = _fedstil_cfg(tmpdir, "single")
+    single = _run_fedstil_rounds(FedContext(), common, exp)
+
+    port = _free_port()
+    mp.spawn(_worker_fedstil, args=(2, port, tmpdir), nprocs=2, join=True)
+    with open(os.path.join(tmpdir, "fedstil_state.pkl"), "rb") as f:
+        dist_state = pickle.load(f)
+    assert set(single) == set(dist_state)
+    for n in single:
+        assert torch.allclose(single[n], dist_state[n], atol=1e-6), n
