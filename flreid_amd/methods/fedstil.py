@@ -67,6 +67,7 @@ class Model(ModelModule):
 
         self.ids = set()
         self.examplars: Dict[int, List] = {}
+        self._egraphs: Dict = {}     # hipGraph cache for eval forwards
 
         # head = earliest stage containing an adaptive leaf
         leaves = adaptive_leaves(self.net)
@@ -85,6 +86,24 @@ class Model(ModelModule):
 
     def pre_trained_module_leaves(self):
         return non_adaptive_leaves(self.net)
+
+    def eval_graphed(self, key: str, fn, data: torch.Tensor, full: bool):
+        """hipGraph-replay an EVAL forward (pure, side-effect-free):
+        weights/BN stats update in place between rounds so one capture is
+        valid for the experiment's lifetime.  Partial batches run eager."""
+        from flreid_amd.runtime.hipgraph import GraphedStep, hipgraph_enabled
+        if (not full or not hipgraph_enabled()
+                or not str(data.device).startswith("cuda")):
+            return fn(data)
+        cache_key = (key, tuple(data.shape))
+        gs = self._egraphs.get(cache_key)
+        if gs is None:
+            gs = GraphedStep(lambda d: (fn(d),))
+            gs.warmup(data)
+            out = gs.capture(data)
+            self._egraphs[cache_key] = gs
+            return out[0]
+        return gs(data)[0]
 
     def head_forward(self, head_input: torch.Tensor):
         """(score, feature) from cached prototype features — the reference's
@@ -126,10 +145,13 @@ class Model(ModelModule):
         stay in HBM."""
         protos, pids, classes, feats = [], [], [], []
         self.eval()
+        bsz = getattr(proto_loader, "batch_size", None)
         for data, person_id, class_id in proto_loader:
             data = data.to(device)
             with autocast(device):
-                _score_feat = self.head_forward(data)
+                _score_feat = self.eval_graphed(
+                    "herd_fwd", self.head_forward, data,
+                    full=data.shape[0] == bsz)
             # train-mode tuple or eval feature — capture the feature part
             feature = _score_feat[1] if isinstance(_score_feat, tuple) else _score_feat
             protos.append(data)
@@ -298,11 +320,14 @@ class Operator(BaseReIDOperator):
         device = model.device
         taps, pids, classes = [], [], []
         model.eval()
+        bsz = getattr(source_loader, "batch_size", None)
         with torch.no_grad():
             for data, person_id, class_id in source_loader:
                 data = data.to(device, non_blocking=True)
                 with autocast(device):
-                    _out, tap = model.tap_forward(data)
+                    tap = model.eval_graphed(
+                        "tap_fwd", lambda d: model.tap_forward(d)[1], data,
+                        full=data.shape[0] == bsz)
                 taps.append(tap.float())
                 pids.append(person_id)
                 classes.append(class_id)
