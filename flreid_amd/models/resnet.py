@@ -22,10 +22,24 @@ from typing import List, Type, Union
 import torch
 import torch.nn as nn
 
+from flreid_amd import ops
 from flreid_amd.tools.logger import Logger
 from flreid_amd.tools.winit import weights_init_classifier, weights_init_kaiming
 
 _log = Logger("models.resnet")
+
+
+class EvalFusedBatchNorm2d(nn.BatchNorm2d):
+    """BatchNorm2d whose EVAL forward runs the fused one-pass HIP kernel on
+    GPU (frozen-backbone prototype capture + validation are the ReID hot
+    eval paths); train mode and CPU fall through to torch."""
+
+    def forward(self, x):
+        if not self.training:
+            out = ops.bn_eval_2d(x, self)
+            if out is not None:
+                return out
+        return super().forward(x)
 
 
 def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
@@ -43,10 +57,10 @@ class BasicBlock(nn.Module):
                  downsample: nn.Module = None):
         super().__init__()
         self.conv1 = conv3x3(cin, planes, stride)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = EvalFusedBatchNorm2d(planes)
         self.relu = nn.ReLU(inplace=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = EvalFusedBatchNorm2d(planes)
         self.downsample = downsample
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -63,11 +77,11 @@ class Bottleneck(nn.Module):
                  downsample: nn.Module = None):
         super().__init__()
         self.conv1 = conv1x1(cin, planes)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = EvalFusedBatchNorm2d(planes)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = EvalFusedBatchNorm2d(planes)
         self.conv3 = conv1x1(planes, planes * self.expansion)
-        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.bn3 = EvalFusedBatchNorm2d(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
 
@@ -87,7 +101,7 @@ class ResNetTrunk(nn.Module):
         super().__init__()
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = EvalFusedBatchNorm2d(64)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_stage(block, 64, layers[0])
@@ -102,7 +116,7 @@ class ResNetTrunk(nn.Module):
         if stride != 1 or self.inplanes != planes * block.expansion:
             downsample = nn.Sequential(
                 conv1x1(self.inplanes, planes * block.expansion, stride),
-                nn.BatchNorm2d(planes * block.expansion),
+                EvalFusedBatchNorm2d(planes * block.expansion),
             )
         blocks = [block(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes * block.expansion

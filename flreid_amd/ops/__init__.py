@@ -412,3 +412,28 @@ def adaptive_linear(x: torch.Tensor, gw: torch.Tensor, atten: torch.Tensor,
         return _AdaptiveLinearFn.apply(x, gw, atten, aw, bias)
     theta = adaptive_compose(gw, atten, aw)
     return torch.nn.functional.linear(x, theta, bias)
+
+
+def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
+    """Fused eval-mode BatchNorm2d (one coalesced pass; MIOpen's inference
+    kernel measured ~0.3 TB/s on ReID shapes).  Returns None when the fused
+    path does not apply (train mode, CPU, missing stats/extension)."""
+    if (not x.is_cuda or x.requires_grad or bn.running_mean is None
+            or x.dtype not in (torch.float32, torch.bfloat16) or x.dim() != 4):
+        return None
+    ext = _load_extension()
+    if ext is None:
+        return None
+    nhwc = x.is_contiguous(memory_format=torch.channels_last)
+    if not nhwc and not x.is_contiguous():
+        x = x.contiguous()
+    out = torch.empty_like(x)
+    n, c, h, w = x.shape
+    ext.bn_eval(x.data_ptr(), out.data_ptr(),
+                bn.weight.detach().float().contiguous().data_ptr(),
+                bn.bias.detach().float().contiguous().data_ptr(),
+                bn.running_mean.float().contiguous().data_ptr(),
+                bn.running_var.float().contiguous().data_ptr(),
+                x.numel(), c, h * w, float(bn.eps), int(nhwc), _dt(x),
+                _stream())
+    return out

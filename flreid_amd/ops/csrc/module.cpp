@@ -39,6 +39,9 @@ extern "C" void flreid_adaptive_linear_fwd(const void*, const float*,
                                            const float*, const float*,
                                            const float*, void*, int, int,
                                            int, int, hipStream_t);
+extern "C" void flreid_bn_eval(const void*, void*, const float*, const float*,
+                               const float*, const float*, int64_t, int,
+                               int64_t, float, int, int, hipStream_t);
 }  // namespace flreid
 
 namespace py = pybind11;
@@ -124,6 +127,16 @@ PYBIND11_MODULE(_flreid_hip, m) {
               (const void*)X, (const float*)GW, (const float*)AW,
               (const float*)ATTEN, (const float*)BIAS, (void*)OUT, M, N, K,
               split_layout, as_stream(stream));
+        });
+
+  m.def("bn_eval",
+        [](uintptr_t x, uintptr_t y, uintptr_t gamma, uintptr_t beta,
+           uintptr_t mean, uintptr_t var, int64_t numel, int C, int64_t HW,
+           float eps, int nhwc, int dtype, uintptr_t stream) {
+          flreid::flreid_bn_eval((const void*)x, (void*)y,
+                                 (const float*)gamma, (const float*)beta,
+                                 (const float*)mean, (const float*)var, numel,
+                                 C, HW, eps, nhwc, dtype, as_stream(stream));
         });
 
   m.def("triplet_bwd",

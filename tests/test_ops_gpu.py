@@ -268,3 +268,21 @@ def test_adaptive_linear_autograd_matches_eager():
     assert torch.allclose(x.grad, x2.grad, atol=5e-3, rtol=5e-2)
     assert torch.allclose(aw.grad, aw2.grad, atol=5e-3, rtol=5e-2)
     assert torch.allclose(bias.grad, bias2.grad, atol=5e-3, rtol=5e-2)
+
+
+def test_bn_eval_fused_matches_torch():
+    import torch.nn as nn
+    bn = nn.BatchNorm2d(32).cuda()
+    bn.running_mean.uniform_(-1, 1)
+    bn.running_var.uniform_(0.5, 2.0)
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.data.uniform_(-1, 1)
+    bn.eval()
+    for fmt in (torch.contiguous_format, torch.channels_last):
+        x = torch.randn(8, 32, 16, 8, device="cuda").to(memory_format=fmt)
+        out = ops.bn_eval_2d(x, bn)
+        expected = bn(x)
+        assert torch.allclose(out, expected, atol=1e-5, rtol=1e-5)
+        xb = x.bfloat16()
+        out_b = ops.bn_eval_2d(xb, bn)
+        assert torch.allclose(out_b.float(), expected, atol=5e-2, rtol=5e-2)
