@@ -203,7 +203,6 @@ extern "C" void flreid_importance(float* F, const void* g, int64_t numel,
   HIP_CHECK(hipGetLastError());
 }
 
-}  // namespace flreid
 
 // --------------------------------------------------------------------------
 // eval-mode BatchNorm2d: y = (x − μ[c])·rsqrt(σ²[c]+eps)·γ[c] + β[c]
@@ -257,3 +256,5 @@ extern "C" void flreid_bn_eval(const void* x, void* y, const float* gamma,
   }
   HIP_CHECK(hipGetLastError());
 }
+
+}  // namespace flreid
