@@ -136,6 +136,12 @@ def main():
     if args.channels_last and device_is_cuda:
         for c in clients:
             c.model.net.to(memory_format=torch.channels_last)
+            # the 7x7 stem stays NCHW: MIOpen's NHWC path falls back to a
+            # naive kernel for it (measured 2.4 ms/call); the two layout
+            # transposes around the stem cost microseconds
+            stem = getattr(getattr(c.model.net, "base", None), "conv1", None)
+            if stem is not None:
+                stem.weight.data = stem.weight.data.contiguous()
 
     def one_round(r):
         stage.process_one_round(r, server, by_name, client_names, exp, log)
