@@ -18,11 +18,20 @@ methods with data-dependent state shapes).
 
 from __future__ import annotations
 
+import os
 from typing import Any, Dict, List, Tuple
 
 import torch
 
 _SENTINEL = "__flreid_tensor__"
+
+
+def _wire_dtype() -> torch.dtype:
+    """Wire format for the flat gather: fp32 (default, bit-transparent) or
+    bf16 (FLREID_COMM_DTYPE=bf16 — halves xGMI traffic at the cost of one
+    bf16 round-trip on the exchanged states)."""
+    return (torch.bfloat16 if os.environ.get("FLREID_COMM_DTYPE", "fp32")
+            == "bf16" else torch.float32)
 
 
 def _flatten(state: Any, path: str, tensors: List[Tuple[str, torch.Tensor]]):
@@ -69,11 +78,12 @@ def sync_client_states(ctx, local_uploads: Dict[str, Any]) -> Dict[str, Any]:
     for cname in sorted(local_uploads.keys()):
         tensors: List[Tuple[str, torch.Tensor]] = []
         skeleton = _flatten(local_uploads[cname], cname, tensors)
+        wire = _wire_dtype()
         if tensors:
-            flat = torch.cat([t.detach().reshape(-1).to(torch.float32)
+            flat = torch.cat([t.detach().reshape(-1).to(wire)
                               for _n, t in tensors]).to(device)
         else:
-            flat = torch.zeros(0, dtype=torch.float32, device=device)
+            flat = torch.zeros(0, dtype=wire, device=device)
         metas[cname] = skeleton
         flats[cname] = flat
 
@@ -88,7 +98,8 @@ def sync_client_states(ctx, local_uploads: Dict[str, Any]) -> Dict[str, Any]:
     if len(strides) == 1 and any(counts):
         stride = next(iter(strides))
         max_clients = max(counts)
-        local = torch.zeros(max_clients * stride, dtype=torch.float32, device=device)
+        local = torch.zeros(max_clients * stride, dtype=_wire_dtype(),
+                            device=device)
         for i, cname in enumerate(sorted(flats.keys())):
             local[i * stride:(i + 1) * stride] = flats[cname]
         gathered = ctx.all_gather_flat(local)          # [W, max_clients*stride]
