@@ -135,7 +135,13 @@ def main():
     by_name = {c.client_name: c for c in clients}
     if args.channels_last and device_is_cuda:
         for c in clients:
-            c.model.net.to(memory_format=torch.channels_last)
+            # per-tensor conversion: methods with >4-D parameters (FedWeIT's
+            # stacked knowledge base) break Module.to(channels_last)
+            for m in c.model.net.modules():
+                for p in list(m.parameters(recurse=False)) + \
+                        list(m.buffers(recurse=False)):
+                    if p.dim() == 4:
+                        p.data = p.data.to(memory_format=torch.channels_last)
             # the 7x7 stem stays NCHW: MIOpen's NHWC path falls back to a
             # naive kernel for it (measured 2.4 ms/call); the two layout
             # transposes around the stem cost microseconds
