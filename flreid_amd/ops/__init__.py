@@ -437,3 +437,21 @@ def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
                 x.numel(), c, h * w, float(bn.eps), int(nhwc), _dt(x),
                 _stream())
     return out
+
+
+def conv3x3_fwd_nhwc(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """Hand-written 3×3 s1 p1 NHWC bf16 conv forward (K1; frozen-backbone
+    eval path).  x: channels-last bf16 [N, C, H, W]; weight: fp32
+    channels-last [K, C, 3, 3].  Returns channels-last bf16 [N, K, H, W]."""
+    ext = _ext_or_raise("conv3x3_fwd")
+    n, c, h, w = x.shape
+    k = weight.shape[0]
+    assert x.is_contiguous(memory_format=torch.channels_last)
+    wcl = weight.detach()
+    if not wcl.is_contiguous(memory_format=torch.channels_last):
+        wcl = wcl.contiguous(memory_format=torch.channels_last)
+    out = torch.empty(n, k, h, w, device=x.device,
+                      dtype=torch.bfloat16).to(memory_format=torch.channels_last)
+    ext.conv3x3_fwd(x.data_ptr(), wcl.float().data_ptr(), out.data_ptr(),
+                    n, h, w, c, k, _stream())
+    return out

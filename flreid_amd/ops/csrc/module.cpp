@@ -42,6 +42,8 @@ extern "C" void flreid_adaptive_linear_fwd(const void*, const float*,
 extern "C" void flreid_bn_eval(const void*, void*, const float*, const float*,
                                const float*, const float*, int64_t, int,
                                int64_t, float, int, int, hipStream_t);
+extern "C" void flreid_conv3x3_fwd(const void*, const float*, void*, int, int,
+                                   int, int, int, hipStream_t);
 }  // namespace flreid
 
 namespace py = pybind11;
@@ -137,6 +139,14 @@ PYBIND11_MODULE(_flreid_hip, m) {
                                  (const float*)gamma, (const float*)beta,
                                  (const float*)mean, (const float*)var, numel,
                                  C, HW, eps, nhwc, dtype, as_stream(stream));
+        });
+
+  m.def("conv3x3_fwd",
+        [](uintptr_t x, uintptr_t w, uintptr_t y, int NB, int H, int Wd,
+           int C, int K, uintptr_t stream) {
+          flreid::flreid_conv3x3_fwd((const void*)x, (const float*)w,
+                                     (void*)y, NB, H, Wd, C, K,
+                                     as_stream(stream));
         });
 
   m.def("triplet_bwd",

@@ -286,3 +286,16 @@ def test_bn_eval_fused_matches_torch():
         xb = x.bfloat16()
         out_b = ops.bn_eval_2d(xb, bn)
         assert torch.allclose(out_b.float(), expected, atol=5e-2, rtol=5e-2)
+
+
+def test_conv3x3_fwd_matches_torch():
+    torch.manual_seed(0)
+    for (n, c, h, w, k) in ((4, 32, 16, 8, 32), (2, 64, 7, 5, 16),
+                            (8, 512, 16, 8, 512)):
+        x = torch.randn(n, c, h, w, device="cuda").bfloat16() \
+            .to(memory_format=torch.channels_last)
+        wt = torch.randn(k, c, 3, 3, device="cuda") * (1.0 / c)
+        out = ops.conv3x3_fwd_nhwc(x, wt)
+        expected = torch.nn.functional.conv2d(x.float(), wt, padding=1)
+        assert torch.allclose(out.float(), expected, atol=0.1, rtol=5e-2), \
+            (n, c, h, w, k, (out.float() - expected).abs().max())
