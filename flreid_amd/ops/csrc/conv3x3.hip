@@ -19,42 +19,42 @@ namespace flreid {
 using cbf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using cf32x4 = __attribute__((ext_vector_type(4))) float;
 
-constexpr int CV_BM = 64;    // output rows per block (4 m-fragments)
-constexpr int CV_BN = 32;    // output channels per block (2 waves)
+constexpr int CV_BM = 128;   // output rows per block (8 m-fragments)
+constexpr int CV_BN = 64;    // output channels per block (4 waves x 16)
 constexpr int CV_BK = 32;    // input-channel tile
 constexpr int CV_PAD = 2;
 
 struct CvPrefetch {
   ushort4 xr[4];
   float4 wr[2];
-  int valid[4];
 };
 
-__global__ __launch_bounds__(128) void conv3x3_fwd_kernel(
+__global__ __launch_bounds__(256) void conv3x3_fwd_kernel(
     const __hip_bfloat16* __restrict__ X, const float* __restrict__ W,
     __hip_bfloat16* __restrict__ Y, int NB, int H, int Wd, int C, int K) {
   __shared__ __hip_bfloat16 lx[2][CV_BM][CV_BK + CV_PAD];
   __shared__ __hip_bfloat16 lw[2][CV_BN][CV_BK + CV_PAD];
+  static_assert(CV_BM == 128 && CV_BN == 64, "geometry assumptions");
 
   const int m0 = blockIdx.x * CV_BM;
   const int k0c = blockIdx.y * CV_BN;        // output-channel block
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;
+  const int wave = tid >> 6;                 // 4 waves x one 16-col fragment
   const int fn = wave * 16 + (lane & 15);
 
   const int M = NB * H * Wd;
   const int lc4 = (tid & 7) * 4;
-  const int lr0 = tid >> 3;                  // 16 rows / pass
+  const int lr0 = tid >> 3;                  // 32 rows / pass
 
-  cf32x4 acc[4] = {{}, {}, {}, {}};
+  cf32x4 acc[8] = {{}, {}, {}, {}, {}, {}, {}, {}};
 
   // decode the 4 output rows this thread stages (shared across shifts)
   int row_h[4], row_w[4];
   int64_t row_base[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
-    const int m = m0 + lr0 + 16 * i;
+    const int m = m0 + lr0 + 32 * i;
     const int hw = m % (H * Wd);
     row_h[i] = hw / Wd;
     row_w[i] = hw % Wd;
@@ -72,18 +72,17 @@ __global__ __launch_bounds__(128) void conv3x3_fwd_kernel(
     const int64_t xoff = ((int64_t)r * Wd + s) * C + ck + lc4;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int m = m0 + lr0 + 16 * i;
+      const int m = m0 + lr0 + 32 * i;
       const int hh = row_h[i] + r;
       const int ww = row_w[i] + s;
       const bool ok = (m < M) && hh >= 0 && hh < H && ww >= 0 && ww < Wd;
-      p.valid[i] = ok;
       p.xr[i] = ok ? *(const ushort4*)(X + row_base[i] + xoff)
                    : ushort4{0, 0, 0, 0};
     }
-    // w[k, r, s, c]: rows k = k0c + lr0 (+16)
+    // w[k, r, s, c]: rows k = k0c + lr0 (+32)
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int kk = k0c + lr0 + 16 * i;
+      const int kk = k0c + lr0 + 32 * i;
       float4 v = {0.f, 0.f, 0.f, 0.f};
       if (kk < K) {
         v = *(const float4*)(W + (((int64_t)kk * 3 + (r + 1)) * 3 + (s + 1)) * C
@@ -96,7 +95,7 @@ __global__ __launch_bounds__(128) void conv3x3_fwd_kernel(
   auto store_tile = [&](const CvPrefetch& p, int buf) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      __hip_bfloat16* dst = &lx[buf][lr0 + 16 * i][lc4];
+      __hip_bfloat16* dst = &lx[buf][lr0 + 32 * i][lc4];
       dst[0] = *(const __hip_bfloat16*)&p.xr[i].x;
       dst[1] = *(const __hip_bfloat16*)&p.xr[i].y;
       dst[2] = *(const __hip_bfloat16*)&p.xr[i].z;
@@ -104,7 +103,7 @@ __global__ __launch_bounds__(128) void conv3x3_fwd_kernel(
     }
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      __hip_bfloat16* dst = &lw[buf][lr0 + 16 * i][lc4];
+      __hip_bfloat16* dst = &lw[buf][lr0 + 32 * i][lc4];
       dst[0] = __float2bfloat16(p.wr[i].x);
       dst[1] = __float2bfloat16(p.wr[i].y);
       dst[2] = __float2bfloat16(p.wr[i].z);
@@ -130,7 +129,7 @@ __global__ __launch_bounds__(128) void conv3x3_fwd_kernel(
       bfrag[e] = *reinterpret_cast<const __bf16*>(&lw[buf][fn][kk]);
     }
 #pragma unroll
-    for (int mf = 0; mf < 4; ++mf) {
+    for (int mf = 0; mf < 8; ++mf) {
       cbf16x8 afrag;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
@@ -151,7 +150,7 @@ __global__ __launch_bounds__(128) void conv3x3_fwd_kernel(
   const int kc = k0c + fn;
   if (kc >= K) return;
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < 8; ++mf) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int mr = m0 + mf * 16 + (lane >> 4) * 4 + reg;
@@ -170,7 +169,7 @@ extern "C" void flreid_conv3x3_fwd(const void* X, const float* W, void* Y,
   }
   const int M = NB * H * Wd;
   dim3 grid((M + CV_BM - 1) / CV_BM, (K + CV_BN - 1) / CV_BN);
-  dim3 block(128);
+  dim3 block(256);
   hipLaunchKernelGGL(conv3x3_fwd_kernel, grid, block, 0, stream,
                      (const __hip_bfloat16*)X, W, (__hip_bfloat16*)Y, NB, H,
                      Wd, C, K);
