@@ -442,10 +442,13 @@ def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
 def conv3x3_fwd_nhwc(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """Hand-written 3×3 s1 p1 NHWC bf16 conv forward (K1; frozen-backbone
     eval path).  x: channels-last bf16 [N, C, H, W]; weight: fp32
-    channels-last [K, C, 3, 3].  Returns channels-last bf16 [N, K, H, W]."""
-    ext = _ext_or_raise("conv3x3_fwd")
+    channels-last [K, C, 3, 3].  Returns channels-last bf16 [N, K, H, W].
+    Falls back to the library conv outside the kernel constraints."""
     n, c, h, w = x.shape
     k = weight.shape[0]
+    if c % 32 != 0 or k % 16 != 0:
+        return torch.nn.functional.conv2d(x, weight.to(x.dtype), padding=1)
+    ext = _ext_or_raise("conv3x3_fwd")
     assert x.is_contiguous(memory_format=torch.channels_last)
     wcl = weight.detach()
     if not wcl.is_contiguous(memory_format=torch.channels_last):
