@@ -52,8 +52,11 @@ def parse_args():
                    help="train images per identity per task")
     p.add_argument("--num-classes", type=int, default=8000)
     p.add_argument("--cpu", action="store_true", help="debug on CPU")
-    p.add_argument("--channels-last", action="store_true",
-                   help="NHWC weights (MIOpen channels-last conv path)")
+    p.add_argument("--channels-last", dest="channels_last",
+                   action="store_true", default=True,
+                   help="NHWC weights (MIOpen channels-last conv path; default)")
+    p.add_argument("--no-channels-last", dest="channels_last",
+                   action="store_false")
     return p.parse_args()
 
 
