@@ -98,6 +98,29 @@ def main():
         results["compose_eager"] = bench(
             lambda: ref.adaptive_compose(gw, atten, aw), n)
 
+    # fused train-mode BN fwd+bwd vs MIOpen (head-epoch shape: M=2048, C=512)
+    import torch.nn as nn
+    for c in (512, 2048):
+        bn = nn.BatchNorm2d(c).cuda().train()
+        xb = (torch.randn(64, c, 8, 4, device="cuda").bfloat16()
+              .to(memory_format=torch.channels_last))
+
+        def fused_bn():
+            x1 = xb.detach().requires_grad_(True)
+            y = ops.bn_train_2d(x1, bn)
+            y.backward(y.detach())
+
+        def miopen_bn():
+            x1 = xb.detach().requires_grad_(True)
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                y = nn.functional.batch_norm(
+                    x1, bn.running_mean, bn.running_var, bn.weight, bn.bias,
+                    True, 0.1, bn.eps)
+            y.backward(y.detach())
+
+        results[f"bn_train_c{c}_fused"] = bench(fused_bn, n)
+        results[f"bn_train_c{c}_miopen"] = bench(miopen_bn, n)
+
     for k in sorted(results):
         v = results[k]
         print(f"{k:28s} {v:10.3f}" + (" ms" if "_TF" not in k and "GBps" not in k else ""))

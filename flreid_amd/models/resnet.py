@@ -30,13 +30,22 @@ _log = Logger("models.resnet")
 
 
 class EvalFusedBatchNorm2d(nn.BatchNorm2d):
-    """BatchNorm2d whose EVAL forward runs the fused one-pass HIP kernel on
-    GPU (frozen-backbone prototype capture + validation are the ReID hot
-    eval paths); train mode and CPU fall through to torch."""
+    """BatchNorm2d with fused HIP paths on GPU:
+      - EVAL: one-pass normalize kernel (prototype capture / validation —
+        the ReID hot eval paths);
+      - TRAIN on small channels-last batches (the FedSTIL head epoch on
+        cached prototype features): single-kernel fwd/bwd with in-kernel
+        statistics and running-stat update, replacing MIOpen's 5-kernel
+        chain + fp32 autocast round-trip per layer (see ops/csrc/bn_train.hip).
+    CPU and out-of-regime shapes fall through to torch."""
 
     def forward(self, x):
         if not self.training:
             out = ops.bn_eval_2d(x, self)
+            if out is not None:
+                return out
+        else:
+            out = ops.bn_train_2d(x, self)
             if out is not None:
                 return out
         return super().forward(x)

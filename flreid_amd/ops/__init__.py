@@ -439,6 +439,94 @@ def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
     return out
 
 
+class _BnTrain2dFn(torch.autograd.Function):
+    """Fused training BatchNorm2d over channels-last rows (bn_train.hip).
+
+    One kernel per direction instead of MIOpen's 5-kernel chains +
+    SubTensorOp casts + the autocast fp32 round-trip; running stats update
+    inside the forward kernel (replay-safe for hipGraph capture)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps):
+        ext = _ext_or_raise("bn_train")
+        n, c, h, w = x.shape
+        m = n * h * w
+        y = torch.empty_like(x)
+        smean = torch.empty(c, device=x.device, dtype=torch.float32)
+        sinv = torch.empty(c, device=x.device, dtype=torch.float32)
+        # split-row partial-sum workspace (overwritten densely every call —
+        # no zeroing kernel needed)
+        nslab = ext.bn_train_nslab(m, c)
+        part = torch.empty(2, nslab, c, device=x.device, dtype=torch.float32)
+        has_rs = running_mean is not None
+        ext.bn_train_fwd(
+            x.data_ptr(), y.data_ptr(),
+            gamma.detach().contiguous().data_ptr(),
+            beta.detach().contiguous().data_ptr(),
+            running_mean.data_ptr() if has_rs else 0,
+            running_var.data_ptr() if has_rs else 0,
+            smean.data_ptr(), sinv.data_ptr(),
+            part[0].data_ptr(), part[1].data_ptr(), m, c,
+            float(momentum), float(eps), float(m) / float(m - 1),
+            _dt(x), _stream())
+        ctx.save_for_backward(x, gamma, smean, sinv)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, smean, sinv = ctx.saved_tensors
+        if not dy.is_contiguous(memory_format=torch.channels_last):
+            dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != x.dtype:
+            dy = dy.to(x.dtype)
+        ext = _ext_or_raise("bn_train")
+        n, c, h, w = x.shape
+        dx = torch.empty_like(x)
+        dgamma = torch.empty(c, device=x.device, dtype=torch.float32)
+        dbeta = torch.empty(c, device=x.device, dtype=torch.float32)
+        m = n * h * w
+        nslab = ext.bn_train_nslab(m, c)
+        part = torch.empty(2, nslab, c, device=x.device, dtype=torch.float32)
+        ext.bn_train_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),
+                         gamma.detach().contiguous().data_ptr(),
+                         smean.data_ptr(), sinv.data_ptr(),
+                         dgamma.data_ptr(), dbeta.data_ptr(),
+                         part[0].data_ptr(), part[1].data_ptr(),
+                         m, c, _dt(x), _stream())
+        return dx, dgamma, dbeta, None, None, None, None
+
+
+def bn_train_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
+    """Fused TRAINING-mode BatchNorm2d for the head-epoch regime.  Applies
+    when x is a CUDA channels-last 4-D tensor with C % 64 == 0 and a small
+    row count (M = N·H·W ≤ 4096 — the cached-prototype batches; large-M
+    full-image training stays on MIOpen's multi-block reduction).  Returns
+    None when the fused path does not apply."""
+    if os.environ.get("FLREID_NO_FUSED_BN", "0") == "1":
+        return None
+    if (not x.is_cuda or x.dim() != 4
+            or x.dtype not in (torch.float32, torch.bfloat16)
+            or bn.weight is None or bn.bias is None
+            or not x.is_contiguous(memory_format=torch.channels_last)):
+        return None
+    n, c, h, w = x.shape
+    m = n * h * w
+    if c % 64 != 0 or m < 2 or m > 4096:
+        return None
+    if bn.momentum is None:  # cumulative-average mode: keep torch semantics
+        return None
+    if not extension_available():
+        return None
+    if bn.track_running_stats and bn.running_mean is not None:
+        if bn.num_batches_tracked is not None:
+            bn.num_batches_tracked.add_(1)
+        rm, rv = bn.running_mean, bn.running_var
+    else:
+        rm = rv = None
+    return _BnTrain2dFn.apply(x, bn.weight, bn.bias, rm, rv,
+                              bn.momentum, bn.eps)
+
+
 def conv3x3_fwd_nhwc(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """Hand-written 3×3 s1 p1 NHWC bf16 conv forward (K1; frozen-backbone
     eval path).  x: channels-last bf16 [N, C, H, W]; weight: fp32

@@ -44,6 +44,15 @@ extern "C" void flreid_bn_eval(const void*, void*, const float*, const float*,
                                int64_t, float, int, int, hipStream_t);
 extern "C" void flreid_conv3x3_fwd(const void*, const float*, void*, int, int,
                                    int, int, int, hipStream_t);
+extern "C" int flreid_bn_train_nslab(int64_t, int);
+extern "C" void flreid_bn_train_fwd(const void*, void*, const float*,
+                                    const float*, float*, float*, float*,
+                                    float*, float*, float*, int64_t, int,
+                                    float, float, float, int, hipStream_t);
+extern "C" void flreid_bn_train_bwd(const void*, const void*, void*,
+                                    const float*, const float*, const float*,
+                                    float*, float*, float*, float*, int64_t,
+                                    int, int, hipStream_t);
 }  // namespace flreid
 
 namespace py = pybind11;
@@ -147,6 +156,38 @@ PYBIND11_MODULE(_flreid_hip, m) {
           flreid::flreid_conv3x3_fwd((const void*)x, (const float*)w,
                                      (void*)y, NB, H, Wd, C, K,
                                      as_stream(stream));
+        });
+
+  m.def("bn_train_nslab", [](int64_t M, int C) {
+    return flreid::flreid_bn_train_nslab(M, C);
+  });
+
+  m.def("bn_train_fwd",
+        [](uintptr_t x, uintptr_t y, uintptr_t gamma, uintptr_t beta,
+           uintptr_t rmean, uintptr_t rvar, uintptr_t smean, uintptr_t sinv,
+           uintptr_t part_a, uintptr_t part_b, int64_t M, int C,
+           float momentum, float eps, float unbiased, int dtype,
+           uintptr_t stream) {
+          flreid::flreid_bn_train_fwd((const void*)x, (void*)y,
+                                      (const float*)gamma, (const float*)beta,
+                                      (float*)rmean, (float*)rvar,
+                                      (float*)smean, (float*)sinv,
+                                      (float*)part_a, (float*)part_b, M, C,
+                                      momentum, eps, unbiased, dtype,
+                                      as_stream(stream));
+        });
+
+  m.def("bn_train_bwd",
+        [](uintptr_t x, uintptr_t dy, uintptr_t dx, uintptr_t gamma,
+           uintptr_t smean, uintptr_t sinv, uintptr_t dgamma, uintptr_t dbeta,
+           uintptr_t part_a, uintptr_t part_b, int64_t M, int C, int dtype,
+           uintptr_t stream) {
+          flreid::flreid_bn_train_bwd((const void*)x, (const void*)dy,
+                                      (void*)dx, (const float*)gamma,
+                                      (const float*)smean, (const float*)sinv,
+                                      (float*)dgamma, (float*)dbeta,
+                                      (float*)part_a, (float*)part_b, M, C,
+                                      dtype, as_stream(stream));
         });
 
   m.def("triplet_bwd",

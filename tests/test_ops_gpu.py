@@ -299,3 +299,66 @@ def test_conv3x3_fwd_matches_torch():
         expected = torch.nn.functional.conv2d(x.float(), wt, padding=1)
         assert torch.allclose(out.float(), expected, atol=0.1, rtol=5e-2), \
             (n, c, h, w, k, (out.float() - expected).abs().max())
+
+
+def test_bn_train_fused_matches_torch():
+    """Fused train-mode BN (fwd stats+normalize+running update, analytic bwd)
+    vs torch.nn.BatchNorm2d in fp32 (ops/csrc/bn_train.hip)."""
+    import copy
+    import torch.nn as nn
+    torch.manual_seed(0)
+    for dtype, atol in ((torch.float32, 1e-4), (torch.bfloat16, 5e-2)):
+        for (n, c, h, w) in ((8, 64, 4, 4), (16, 128, 8, 4), (4, 2048, 8, 4)):
+            bn = nn.BatchNorm2d(c).cuda()
+            bn.weight.data.uniform_(0.5, 1.5)
+            bn.bias.data.uniform_(-1, 1)
+            bn.running_mean.uniform_(-1, 1)
+            bn.running_var.uniform_(0.5, 2.0)
+            bn_ref = copy.deepcopy(bn)
+            bn.train(), bn_ref.train()
+
+            x = (torch.randn(n, c, h, w, device="cuda")
+                 .to(dtype).to(memory_format=torch.channels_last)
+                 .requires_grad_(True))
+            x_ref = x.detach().float().clone().requires_grad_(True)
+
+            y = ops.bn_train_2d(x, bn)
+            assert y is not None, "fused path did not engage"
+            assert y.is_contiguous(memory_format=torch.channels_last)
+            y_ref = bn_ref(x_ref)
+            assert torch.allclose(y.float(), y_ref, atol=atol, rtol=5e-2)
+
+            # running stats + counter updated like torch
+            assert torch.allclose(bn.running_mean, bn_ref.running_mean,
+                                  atol=atol, rtol=1e-2)
+            assert torch.allclose(bn.running_var, bn_ref.running_var,
+                                  atol=atol, rtol=1e-2)
+            assert int(bn.num_batches_tracked) == int(bn_ref.num_batches_tracked)
+
+            dy = torch.randn_like(y_ref)
+            y.backward(dy.to(dtype))
+            y_ref.backward(dy)
+            assert torch.allclose(x.grad.float(), x_ref.grad,
+                                  atol=atol * 10, rtol=5e-2), \
+                (dtype, c, (x.grad.float() - x_ref.grad).abs().max())
+            assert torch.allclose(bn.weight.grad, bn_ref.weight.grad,
+                                  atol=atol * 10, rtol=5e-2)
+            assert torch.allclose(bn.bias.grad, bn_ref.bias.grad,
+                                  atol=atol * 10, rtol=5e-2)
+
+
+def test_bn_train_fused_declines_out_of_regime():
+    import torch.nn as nn
+    bn = nn.BatchNorm2d(64).cuda().train()
+    # NCHW-contiguous input -> decline
+    x = torch.randn(8, 64, 4, 4, device="cuda")
+    assert ops.bn_train_2d(x, bn) is None
+    # big M (full-image training regime) -> decline
+    xl = torch.randn(64, 64, 32, 16, device="cuda") \
+        .to(memory_format=torch.channels_last)
+    assert ops.bn_train_2d(xl, bn) is None
+    # C % 64 != 0 -> decline
+    bn2 = nn.BatchNorm2d(48).cuda().train()
+    x2 = torch.randn(8, 48, 4, 4, device="cuda") \
+        .to(memory_format=torch.channels_last)
+    assert ops.bn_train_2d(x2, bn2) is None
