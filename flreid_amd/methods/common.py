@@ -9,6 +9,7 @@ distinctive pieces (penalties, state schemas, aggregation).
 from __future__ import annotations
 
 import collections
+import os
 from typing import Any, Dict, List, Optional, Union
 
 import torch
@@ -64,6 +65,13 @@ class BaseReIDOperator(OperatorModule):
         group = {"params": params, **defaults}
         if capturable:
             group["capturable"] = True
+        # single fused multi-tensor Adam step (one kernel instead of ~9
+        # _multi_tensor passes per step); device-state, hipGraph-capturable
+        if (isinstance(self.optimizer, (torch.optim.Adam, torch.optim.AdamW))
+                and params and params[0].is_cuda
+                and os.environ.get("FLREID_FUSED_ADAM", "1") == "1"):
+            group["fused"] = True
+            group["foreach"] = False
         self.optimizer.param_groups = [group]
 
     # hook: extra loss terms (EWC/MAS penalty, FedProx prox, FedSTIL L1 ...)
