@@ -353,7 +353,11 @@ class Operator(BaseReIDOperator):
         """One head-training step — eager body AND the hipGraph-captured fn
         (must stay replay-safe: fixed shapes, no host syncs)."""
         device = model.device
-        self.optimizer.zero_grad(set_to_none=False)
+        # set_to_none also under graph capture: backward's AccumulateGrad
+        # allocates each .grad from the capture pool exactly once, so replays
+        # rewrite the same buffers — and the ~70 fill/accumulate launches per
+        # step that set_to_none=False costs disappear from the graph
+        self.optimizer.zero_grad(set_to_none=True)
         with autocast(device):
             score, feature = model.head_forward(data)
             loss = 0.0

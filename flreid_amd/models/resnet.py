@@ -51,6 +51,23 @@ class EvalFusedBatchNorm2d(nn.BatchNorm2d):
         return super().forward(x)
 
 
+class EvalFusedBatchNorm1d(nn.BatchNorm1d):
+    """BatchNorm1d (the BNNeck bottleneck) with the same fused HIP train and
+    eval paths as EvalFusedBatchNorm2d — a contiguous [B, C] tensor is one
+    NHWC row block with H·W = 1."""
+
+    def forward(self, x):
+        if not self.training:
+            out = ops.bn_eval_2d(x, self)
+            if out is not None:
+                return out
+        else:
+            out = ops.bn_train_2d(x, self)
+            if out is not None:
+                return out
+        return super().forward(x)
+
+
 def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
@@ -174,7 +191,7 @@ class ResNetReID(nn.Module):
         if neck == "no":
             self.classifier = nn.Linear(self.in_planes, num_classes)
         elif neck == "bnneck":
-            self.bottleneck = nn.BatchNorm1d(self.in_planes)
+            self.bottleneck = EvalFusedBatchNorm1d(self.in_planes)
             self.bottleneck.bias.requires_grad_(False)
             self.classifier = nn.Linear(self.in_planes, num_classes, bias=False)
             self.bottleneck.apply(weights_init_kaiming)

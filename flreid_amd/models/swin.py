@@ -287,7 +287,8 @@ class SwinReID(nn.Module):
         if neck == "no":
             self.classifier = nn.Linear(self.in_planes, num_classes)
         elif neck == "bnneck":
-            self.bottleneck = nn.BatchNorm1d(self.in_planes)
+            from flreid_amd.models.resnet import EvalFusedBatchNorm1d
+            self.bottleneck = EvalFusedBatchNorm1d(self.in_planes)
             self.bottleneck.bias.requires_grad_(False)
             self.classifier = nn.Linear(self.in_planes, num_classes, bias=False)
             self.bottleneck.apply(weights_init_kaiming)

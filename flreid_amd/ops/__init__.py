@@ -419,22 +419,29 @@ def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
     kernel measured ~0.3 TB/s on ReID shapes).  Returns None when the fused
     path does not apply (train mode, CPU, missing stats/extension)."""
     if (not x.is_cuda or x.requires_grad or bn.running_mean is None
-            or x.dtype not in (torch.float32, torch.bfloat16) or x.dim() != 4):
+            or x.dtype not in (torch.float32, torch.bfloat16)
+            or x.dim() not in (2, 4)):
         return None
     ext = _load_extension()
     if ext is None:
         return None
-    nhwc = x.is_contiguous(memory_format=torch.channels_last)
-    if not nhwc and not x.is_contiguous():
-        x = x.contiguous()
+    if x.dim() == 2:      # BatchNorm1d rows: NHWC layout with H·W == 1
+        if not x.is_contiguous():
+            x = x.contiguous()
+        nhwc, hw = True, 1
+    else:
+        nhwc = x.is_contiguous(memory_format=torch.channels_last)
+        if not nhwc and not x.is_contiguous():
+            x = x.contiguous()
+        hw = x.shape[2] * x.shape[3]
     out = torch.empty_like(x)
-    n, c, h, w = x.shape
+    c = x.shape[1]
     ext.bn_eval(x.data_ptr(), out.data_ptr(),
                 bn.weight.detach().float().contiguous().data_ptr(),
                 bn.bias.detach().float().contiguous().data_ptr(),
                 bn.running_mean.float().contiguous().data_ptr(),
                 bn.running_var.float().contiguous().data_ptr(),
-                x.numel(), c, h * w, float(bn.eps), int(nhwc), _dt(x),
+                x.numel(), c, hw, float(bn.eps), int(nhwc), _dt(x),
                 _stream())
     return out
 
@@ -447,10 +454,11 @@ class _BnTrain2dFn(torch.autograd.Function):
     inside the forward kernel (replay-safe for hipGraph capture)."""
 
     @staticmethod
-    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps):
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
+                nbt):
         ext = _ext_or_raise("bn_train")
-        n, c, h, w = x.shape
-        m = n * h * w
+        c = x.shape[1]
+        m = x.numel() // c
         y = torch.empty_like(x)
         smean = torch.empty(c, device=x.device, dtype=torch.float32)
         sinv = torch.empty(c, device=x.device, dtype=torch.float32)
@@ -466,7 +474,8 @@ class _BnTrain2dFn(torch.autograd.Function):
             running_mean.data_ptr() if has_rs else 0,
             running_var.data_ptr() if has_rs else 0,
             smean.data_ptr(), sinv.data_ptr(),
-            part[0].data_ptr(), part[1].data_ptr(), m, c,
+            part[0].data_ptr(), part[1].data_ptr(),
+            nbt.data_ptr() if nbt is not None else 0, m, c,
             float(momentum), float(eps), float(m) / float(m - 1),
             _dt(x), _stream())
         ctx.save_for_backward(x, gamma, smean, sinv)
@@ -475,16 +484,19 @@ class _BnTrain2dFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, gamma, smean, sinv = ctx.saved_tensors
-        if not dy.is_contiguous(memory_format=torch.channels_last):
-            dy = dy.contiguous(memory_format=torch.channels_last)
+        if x.dim() == 4:
+            if not dy.is_contiguous(memory_format=torch.channels_last):
+                dy = dy.contiguous(memory_format=torch.channels_last)
+        elif not dy.is_contiguous():
+            dy = dy.contiguous()
         if dy.dtype != x.dtype:
             dy = dy.to(x.dtype)
         ext = _ext_or_raise("bn_train")
-        n, c, h, w = x.shape
+        c = x.shape[1]
+        m = x.numel() // c
         dx = torch.empty_like(x)
         dgamma = torch.empty(c, device=x.device, dtype=torch.float32)
         dbeta = torch.empty(c, device=x.device, dtype=torch.float32)
-        m = n * h * w
         nslab = ext.bn_train_nslab(m, c)
         part = torch.empty(2, nslab, c, device=x.device, dtype=torch.float32)
         ext.bn_train_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),
@@ -493,38 +505,46 @@ class _BnTrain2dFn(torch.autograd.Function):
                          dgamma.data_ptr(), dbeta.data_ptr(),
                          part[0].data_ptr(), part[1].data_ptr(),
                          m, c, _dt(x), _stream())
-        return dx, dgamma, dbeta, None, None, None, None
+        return dx, dgamma, dbeta, None, None, None, None, None
 
 
 def bn_train_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
-    """Fused TRAINING-mode BatchNorm2d for the head-epoch regime.  Applies
-    when x is a CUDA channels-last 4-D tensor with C % 64 == 0 and a small
-    row count (M = N·H·W ≤ 4096 — the cached-prototype batches; large-M
-    full-image training stays on MIOpen's multi-block reduction).  Returns
+    """Fused TRAINING-mode BatchNorm for the head-epoch regime.  Applies to
+    CUDA channels-last 4-D tensors (BatchNorm2d) and contiguous 2-D tensors
+    (BatchNorm1d — the BNNeck bottleneck) with C % 64 == 0 and a small row
+    count (M = N·H·W ≤ 4096 — the cached-prototype batches; large-M
+    full-image training stays on MIOpen's multi-block reduction).  The
+    num_batches_tracked increment is fused into the forward kernel.  Returns
     None when the fused path does not apply."""
     if os.environ.get("FLREID_NO_FUSED_BN", "0") == "1":
         return None
-    if (not x.is_cuda or x.dim() != 4
-            or x.dtype not in (torch.float32, torch.bfloat16)
-            or bn.weight is None or bn.bias is None
-            or not x.is_contiguous(memory_format=torch.channels_last)):
+    if (not x.is_cuda or x.dtype not in (torch.float32, torch.bfloat16)
+            or bn.weight is None or bn.bias is None):
         return None
-    n, c, h, w = x.shape
-    m = n * h * w
+    if x.dim() == 4:
+        if not x.is_contiguous(memory_format=torch.channels_last):
+            return None
+    elif x.dim() == 2:
+        if not x.is_contiguous():
+            return None
+    else:
+        return None
+    c = x.shape[1]
+    m = x.numel() // c
     if c % 64 != 0 or m < 2 or m > 4096:
         return None
     if bn.momentum is None:  # cumulative-average mode: keep torch semantics
         return None
     if not extension_available():
         return None
+    nbt = None
     if bn.track_running_stats and bn.running_mean is not None:
-        if bn.num_batches_tracked is not None:
-            bn.num_batches_tracked.add_(1)
+        nbt = bn.num_batches_tracked
         rm, rv = bn.running_mean, bn.running_var
     else:
         rm = rv = None
     return _BnTrain2dFn.apply(x, bn.weight, bn.bias, rm, rv,
-                              bn.momentum, bn.eps)
+                              bn.momentum, bn.eps, nbt)
 
 
 def conv3x3_fwd_nhwc(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:

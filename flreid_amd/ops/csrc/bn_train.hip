@@ -151,7 +151,8 @@ __global__ __launch_bounds__(256) void bn_finalize_fwd_kernel(
     float* __restrict__ rmean, float* __restrict__ rvar,
     float* __restrict__ smean, float* __restrict__ sinv,
     const float* __restrict__ part_a, const float* __restrict__ part_b,
-    int64_t M, int C, float momentum, float eps, float unbiased) {
+    unsigned long long* __restrict__ nbt, int64_t M, int C, float momentum,
+    float eps, float unbiased) {
   const int c0 = blockIdx.x * BN_CG;
   const int q = threadIdx.x & 15;
   const int r = threadIdx.x >> 4;
@@ -159,6 +160,12 @@ __global__ __launch_bounds__(256) void bn_finalize_fwd_kernel(
   const int S = gridDim.y * BN_STRIPES;
   const int g = blockIdx.y * BN_STRIPES + r;
   const int nslab = gridDim.y;
+
+  // num_batches_tracked update fused in (torch increments it per train call)
+  if (nbt != nullptr && blockIdx.x == 0 && blockIdx.y == 0 &&
+      threadIdx.x == 0) {
+    ++(*nbt);
+  }
 
   __shared__ float lmean[BN_CG], linv[BN_CG];
   if (threadIdx.x < BN_CG) {
@@ -277,10 +284,10 @@ extern "C" int flreid_bn_train_nslab(int64_t M, int C) {
 extern "C" void flreid_bn_train_fwd(const void* X, void* Y, const float* gamma,
                                     const float* beta, float* rmean,
                                     float* rvar, float* smean, float* sinv,
-                                    float* part_a, float* part_b, int64_t M,
-                                    int C, float momentum, float eps,
-                                    float unbiased, int dtype,
-                                    hipStream_t stream) {
+                                    float* part_a, float* part_b,
+                                    unsigned long long* nbt, int64_t M, int C,
+                                    float momentum, float eps, float unbiased,
+                                    int dtype, hipStream_t stream) {
   if (C % BN_CG != 0) throw std::runtime_error("bn_train: C % 64 != 0");
   dim3 grid(C / BN_CG, bn_nslab(M, C)), block(256);
   if (dtype == kBF16) {
@@ -290,14 +297,14 @@ extern "C" void flreid_bn_train_fwd(const void* X, void* Y, const float* gamma,
     hipLaunchKernelGGL(bn_finalize_fwd_kernel<__hip_bfloat16>, grid, block, 0,
                        stream, (const __hip_bfloat16*)X, (__hip_bfloat16*)Y,
                        gamma, beta, rmean, rvar, smean, sinv, part_a, part_b,
-                       M, C, momentum, eps, unbiased);
+                       nbt, M, C, momentum, eps, unbiased);
   } else {
     hipLaunchKernelGGL((bn_partial_kernel<float, false>), grid, block, 0,
                        stream, (const float*)X, nullptr, nullptr, nullptr,
                        part_a, part_b, M, C);
     hipLaunchKernelGGL(bn_finalize_fwd_kernel<float>, grid, block, 0, stream,
                        (const float*)X, (float*)Y, gamma, beta, rmean, rvar,
-                       smean, sinv, part_a, part_b, M, C, momentum, eps,
+                       smean, sinv, part_a, part_b, nbt, M, C, momentum, eps,
                        unbiased);
   }
   HIP_CHECK(hipGetLastError());

@@ -47,8 +47,9 @@ extern "C" void flreid_conv3x3_fwd(const void*, const float*, void*, int, int,
 extern "C" int flreid_bn_train_nslab(int64_t, int);
 extern "C" void flreid_bn_train_fwd(const void*, void*, const float*,
                                     const float*, float*, float*, float*,
-                                    float*, float*, float*, int64_t, int,
-                                    float, float, float, int, hipStream_t);
+                                    float*, float*, float*,
+                                    unsigned long long*, int64_t, int, float,
+                                    float, float, int, hipStream_t);
 extern "C" void flreid_bn_train_bwd(const void*, const void*, void*,
                                     const float*, const float*, const float*,
                                     float*, float*, float*, float*, int64_t,
@@ -165,14 +166,15 @@ PYBIND11_MODULE(_flreid_hip, m) {
   m.def("bn_train_fwd",
         [](uintptr_t x, uintptr_t y, uintptr_t gamma, uintptr_t beta,
            uintptr_t rmean, uintptr_t rvar, uintptr_t smean, uintptr_t sinv,
-           uintptr_t part_a, uintptr_t part_b, int64_t M, int C,
-           float momentum, float eps, float unbiased, int dtype,
+           uintptr_t part_a, uintptr_t part_b, uintptr_t nbt, int64_t M,
+           int C, float momentum, float eps, float unbiased, int dtype,
            uintptr_t stream) {
           flreid::flreid_bn_train_fwd((const void*)x, (void*)y,
                                       (const float*)gamma, (const float*)beta,
                                       (float*)rmean, (float*)rvar,
                                       (float*)smean, (float*)sinv,
-                                      (float*)part_a, (float*)part_b, M, C,
+                                      (float*)part_a, (float*)part_b,
+                                      (unsigned long long*)nbt, M, C,
                                       momentum, eps, unbiased, dtype,
                                       as_stream(stream));
         });

@@ -362,3 +362,29 @@ def test_bn_train_fused_declines_out_of_regime():
     x2 = torch.randn(8, 48, 4, 4, device="cuda") \
         .to(memory_format=torch.channels_last)
     assert ops.bn_train_2d(x2, bn2) is None
+
+
+def test_bn_train_fused_1d_matches_torch():
+    """2-D input path (BNNeck BatchNorm1d) of the fused train BN."""
+    import copy
+    import torch.nn as nn
+    torch.manual_seed(1)
+    bn = nn.BatchNorm1d(2048).cuda().train()
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.requires_grad_(False)
+    bn_ref = copy.deepcopy(bn)
+    x = torch.randn(64, 2048, device="cuda", requires_grad=True)
+    x_ref = x.detach().clone().requires_grad_(True)
+    y = ops.bn_train_2d(x, bn)
+    assert y is not None
+    y_ref = bn_ref(x_ref)
+    assert torch.allclose(y, y_ref, atol=1e-4, rtol=1e-3)
+    assert torch.allclose(bn.running_var, bn_ref.running_var, atol=1e-4,
+                          rtol=1e-3)
+    assert int(bn.num_batches_tracked) == 1
+    dy = torch.randn_like(y_ref)
+    y.backward(dy)
+    y_ref.backward(dy)
+    assert torch.allclose(x.grad, x_ref.grad, atol=1e-3, rtol=1e-2)
+    assert torch.allclose(bn.weight.grad, bn_ref.weight.grad, atol=1e-3,
+                          rtol=1e-2)
