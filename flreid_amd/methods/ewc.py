@@ -52,6 +52,8 @@ class _LocalLifelongClient(BaseReIDClient):
         self.model.update_model(model_dict)
 
     def save_model(self, model_name: str) -> None:
+        if self._ckpt_disabled():
+            return
         self.save_state(model_name, self.model.model_state(), True)
 
     def update_by_incremental_state(self, state: Dict, **kwargs) -> Any:

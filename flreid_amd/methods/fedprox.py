@@ -79,6 +79,8 @@ class Client(BaseReIDClient):
         self.model.update_model(model_dict)
 
     def save_model(self, model_name: str) -> None:
+        if self._ckpt_disabled():
+            return
         self.save_state(model_name, self.model.model_state(), True)
 
     def get_incremental_state(self, **kwargs) -> Dict:

@@ -486,6 +486,11 @@ class Client(BaseReIDClient):
                 self.load_state(f"{model_name}_examplars", {}))
 
     def save_model(self, model_name: str) -> None:
+        # gate BEFORE building the ckpt payload: the full-state clone and the
+        # per-exemplar .cpu() loop (~2000 small D2H copies per round at
+        # λ_k=2000) must not run when the audit trail is off
+        if self._ckpt_disabled():
+            return
         self.save_state(model_name, self.model.model_state(), True)
         self.save_state(f"{model_name}_examplars",
                         self._examplars_to_ckpt(self.model.examplars), True)
@@ -600,6 +605,9 @@ class Server(ServerModule):
         self.model.update_model(self.load_state(model_name, self.model.model_state()))
 
     def save_model(self, model_name: str) -> None:
+        import os as _os
+        if _os.environ.get("FLREID_DISABLE_CKPT", "0") == "1":
+            return
         self.save_state(model_name, self.model.model_state(), True)
 
     def calculate(self) -> Any:

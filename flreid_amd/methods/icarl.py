@@ -243,6 +243,8 @@ class Client(BaseReIDClient):
             self.model.update_model(self.load_state(model_name, None))
 
     def save_model(self, model_name: str) -> None:
+        if self._ckpt_disabled():
+            return
         self.save_state(model_name, self.model.model_state(), True)
 
     def update_by_incremental_state(self, state: Dict, **kwargs) -> Any:
