@@ -143,72 +143,110 @@ class Model(ModelModule):
         iteratively pick argmin‖μ − (f + Σ picked)/(i+1)‖ (repeats allowed,
         matching the reference's selection loop).  Runs on-device; exemplars
         stay in HBM."""
-        protos, pids, classes, feats = [], [], [], []
         self.eval()
-        bsz = getattr(proto_loader, "batch_size", None)
-        for data, person_id, class_id in proto_loader:
-            data = data.to(device)
-            with autocast(device):
-                _score_feat = self.eval_graphed(
-                    "herd_fwd", self.head_forward, data,
-                    full=data.shape[0] == bsz)
-            # train-mode tuple or eval feature — capture the feature part
-            feature = _score_feat[1] if isinstance(_score_feat, tuple) else _score_feat
-            protos.append(data)
-            pids.append(person_id)
-            classes.append(class_id)
-            feats.append(feature.float())
-        if not protos:
-            return
-        protos = torch.cat(protos)
-        pids = torch.cat(pids)
-        classes = torch.cat(classes)
-        feats = torch.cat(feats)
+        stacked = getattr(proto_loader, "data", None)
+        if torch.is_tensor(stacked):
+            # device-resident TensorBatches: skip the 64-sample batching and
+            # run the (eval-mode, batch-size-invariant) feature forward in
+            # large chunks — 5 launches of the head instead of 40
+            protos = stacked.to(device)
+            if protos.shape[0] == 0:
+                return
+            pids = proto_loader.pids
+            classes = proto_loader.classes
+            feats = []
+            chunk = 512
+            for i in range(0, protos.shape[0], chunk):
+                data = protos[i:i + chunk]
+                with autocast(device):
+                    sf = self.eval_graphed("herd_fwd", self.head_forward, data,
+                                           full=data.shape[0] == chunk)
+                feats.append((sf[1] if isinstance(sf, tuple) else sf).float())
+            feats = torch.cat(feats)
+        else:
+            protos, pids, classes, feats = [], [], [], []
+            bsz = getattr(proto_loader, "batch_size", None)
+            for data, person_id, class_id in proto_loader:
+                data = data.to(device)
+                with autocast(device):
+                    _score_feat = self.eval_graphed(
+                        "herd_fwd", self.head_forward, data,
+                        full=data.shape[0] == bsz)
+                # train-mode tuple or eval feature — capture the feature part
+                feature = _score_feat[1] if isinstance(_score_feat, tuple) \
+                    else _score_feat
+                protos.append(data)
+                pids.append(person_id)
+                classes.append(class_id)
+                feats.append(feature.float())
+            if not protos:
+                return
+            protos = torch.cat(protos)
+            pids = torch.cat(pids)
+            classes = torch.cat(classes)
+            feats = torch.cat(feats)
 
         if person_ids is not None and len(person_ids):
-            keep = torch.tensor([int(p) in set(int(x) for x in person_ids)
-                                 for p in pids], dtype=torch.bool)
+            allowed = torch.as_tensor([int(x) for x in person_ids],
+                                      dtype=pids.dtype)
+            keep = torch.isin(pids, allowed)
             protos, pids, classes, feats = protos[keep.to(protos.device)], \
                 pids[keep], classes[keep], feats[keep.to(feats.device)]
 
         # batched sync-free herding: all identities advance together, padded
-        # to the largest per-identity sample count; 4 kernels per herding step
-        # instead of 4 × n_identities (measured 8k+ micro-launches/round on
-        # the per-identity loop)
-        persons = torch.unique(pids).tolist()
-        per_idx = {p: (pids == p).nonzero(as_tuple=True)[0] for p in persons}
-        nmax = max(ix.numel() for ix in per_idx.values())
-        P = len(persons)
+        # to the largest per-identity sample count; a handful of kernels per
+        # herding step instead of 4 × n_identities.  The per-identity
+        # partition is built by ONE stable sort (no (pids==p).nonzero loop),
+        # and the argmin uses the expanded form
+        #   ‖μ·(i+1) − acc − f‖² = ‖f‖² − 2⟨f, g_i⟩ + const
+        # so no [P, nmax, D] candidate tensor is ever materialised.
         dev = feats.device
+        order = torch.argsort(pids, stable=True)                 # cpu
+        persons_t, counts_t = torch.unique_consecutive(
+            pids[order], return_counts=True)
+        persons = persons_t.tolist()
+        P = len(persons)
+        nmax = int(counts_t.max())
         D = feats.shape[1]
+        offsets = torch.cat([torch.zeros(1, dtype=torch.long),
+                             counts_t.cumsum(0)[:-1]])
+        # position of each (sorted) sample within its identity group
+        pos = torch.arange(pids.numel()) - offsets.repeat_interleave(counts_t)
+        group = torch.arange(P).repeat_interleave(counts_t)
+
+        order_dev = order.to(dev)
         f_pad = torch.zeros(P, nmax, D, device=dev)
-        pad_penalty = torch.full((P, nmax), 0.0, device=dev)
-        for r, p in enumerate(persons):
-            ix = per_idx[p].to(dev)
-            f_pad[r, :ix.numel()] = feats.index_select(0, ix)
-            if ix.numel() < nmax:
-                pad_penalty[r, ix.numel():] = 1e30
-        counts = torch.tensor([per_idx[p].numel() for p in persons],
-                              device=dev, dtype=torch.float32)
+        f_pad[group.to(dev), pos.to(dev)] = feats.index_select(0, order_dev)
+        pad_penalty = torch.zeros(P, nmax, device=dev)
+        pad_mask = pos.new_zeros(P, nmax, dtype=torch.bool)
+        pad_mask[group, pos] = True
+        pad_penalty.masked_fill_(~pad_mask.to(dev), 1e30)
+        counts = counts_t.to(dev, torch.float32)
         mu = f_pad.sum(dim=1) / counts.unsqueeze(1)              # [P, D]
 
+        fsq = (f_pad * f_pad).sum(dim=2)                         # [P, nmax]
         acc = torch.zeros(P, D, device=dev)
         arange_p = torch.arange(P, device=dev)
         step_idx = []
         for i in range(self.m):
-            cand = mu.unsqueeze(1) - (f_pad + acc.unsqueeze(1)) / (i + 1)
-            norms = torch.linalg.vector_norm(cand, dim=2) + pad_penalty
-            idx = norms.argmin(dim=1)                            # [P]
+            g = mu * (i + 1) - acc                               # [P, D]
+            dots = torch.bmm(f_pad, g.unsqueeze(2)).squeeze(2)   # [P, nmax]
+            score = fsq - 2.0 * dots + pad_penalty
+            idx = score.argmin(dim=1)                            # [P]
             step_idx.append(idx)
             acc = acc + f_pad[arange_p, idx]
         sel = torch.stack(step_idx, dim=1).cpu()                 # [P, m]
 
+        # map padded positions back to original sample indices and gather ALL
+        # identities' exemplars with one index_select; per-identity entries
+        # are views into the stacked result (stays HBM-resident)
+        chosen = order.view(-1)[(offsets.unsqueeze(1) + sel).clamp_max(
+            pids.numel() - 1).view(-1)].view(P, -1)
+        proto_sel = protos.index_select(
+            0, chosen.view(-1).to(protos.device)).view(P, -1, *protos.shape[1:])
+        class_sel = classes.index_select(0, chosen.view(-1)).view(P, -1)
         for r, p in enumerate(persons):
-            local = per_idx[p]                                   # cpu indices
-            chosen = local[sel[r]]                               # [m] cpu
-            self.examplars[int(p)] = (
-                protos.index_select(0, chosen.to(protos.device)).clone(),
-                classes.index_select(0, chosen).clone())
+            self.examplars[int(p)] = (proto_sel[r], class_sel[r])
 
     def reduce_examplars(self) -> None:
         for k in list(self.examplars):
