@@ -242,8 +242,8 @@ def importance_update(importance: Dict[str, torch.Tensor],
 kl_distance = ref.kl_distance
 kd_loss = ref.kd_loss
 quadratic_penalty = ref.quadratic_penalty
-l1_drift = ref.l1_drift
 l1_drift_fused = ref.l1_drift_fused
+# l1_drift is defined below: fused HIP multi-tensor path on GPU
 
 
 class _TripletHardFn(torch.autograd.Function):
@@ -444,6 +444,85 @@ def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
                 x.numel(), c, hw, float(bn.eps), int(nhwc), _dt(x),
                 _stream())
     return out
+
+
+_DRIFT_CHUNK = 1 << 16
+_DRIFT_META: Dict = {}
+
+
+class _L1DriftHipFn(torch.autograd.Function):
+    """Fused multi-tensor Σ|p − p₀| (ops/csrc/drift.hip): one read-only pass
+    forward, one sign-write pass backward into a single flat buffer split
+    into per-tensor grad views."""
+
+    @staticmethod
+    def forward(ctx, meta, n_params, *tensors):
+        ext = _ext_or_raise("drift")
+        ptrs, chunks, sizes, total = meta
+        partials = torch.empty(chunks.shape[0], device=tensors[0].device,
+                               dtype=torch.float32)
+        ext.drift_fwd(ptrs.data_ptr(), chunks.data_ptr(), partials.data_ptr(),
+                      chunks.shape[0], _stream())
+        ctx.meta = meta
+        ctx.n_params = n_params
+        ctx.shapes = [t.shape for t in tensors[:n_params]]
+        return partials.sum()
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        ext = _ext_or_raise("drift")
+        ptrs, chunks, sizes, total = ctx.meta
+        flat = torch.empty(total, device=grad_out.device, dtype=torch.float32)
+        g = grad_out.to(torch.float32).contiguous()
+        ext.drift_bwd(ptrs.data_ptr(), chunks.data_ptr(), g.data_ptr(),
+                      flat.data_ptr(), chunks.shape[0], _stream())
+        views = [v.view(s) for v, s in zip(flat.split(sizes), ctx.shapes)]
+        return (None, None, *views, *([None] * ctx.n_params))
+
+
+def _drift_meta(pairs):
+    """(ptr_table, chunk_table, sizes, total) on device, cached on the
+    parameter storages (in-place per-round updates keep pointers stable;
+    a dispatch that rebinds storage changes the key and rebuilds)."""
+    key = tuple((p.data_ptr(), a.data_ptr(), p.numel()) for p, a in pairs)
+    meta = _DRIFT_META.get(key)
+    if meta is not None:
+        return meta
+    dev = pairs[0][0].device
+    ptr_rows, chunk_rows, sizes = [], [], []
+    flat_off = 0
+    for i, (p, a) in enumerate(pairs):
+        ptr_rows.append((p.data_ptr(), a.data_ptr()))
+        n = p.numel()
+        sizes.append(n)
+        off = 0
+        while off < n:
+            ln = min(_DRIFT_CHUNK, n - off)
+            chunk_rows.append((i, off, ln, flat_off + off))
+            off += ln
+        flat_off += n
+    ptrs = torch.tensor(ptr_rows, dtype=torch.int64, device=dev)
+    chunks = torch.tensor(chunk_rows, dtype=torch.int32, device=dev)
+    # entries are never evicted: a captured hipGraph may hold the table
+    # pointers for its lifetime, and each entry is only ~KBs
+    meta = (ptrs, chunks, sizes, flat_off)
+    _DRIFT_META[key] = meta
+    return meta
+
+
+def l1_drift(pairs) -> torch.Tensor:
+    """Σ |p − p₀| over drift pairs — fused HIP multi-tensor path on GPU
+    (fp32, contiguous), _foreach fallback elsewhere."""
+    pairs = [(p, a.detach()) for p, a in pairs]
+    if (pairs and pairs[0][0].is_cuda and extension_available()
+            and all(p.dtype == torch.float32 and p.is_contiguous()
+                    and a.dtype == torch.float32 and a.is_contiguous()
+                    and p.numel() == a.numel() for p, a in pairs)):
+        meta = _drift_meta(pairs)
+        params = [p for p, _ in pairs]
+        anchors = [a for _, a in pairs]
+        return _L1DriftHipFn.apply(meta, len(params), *params, *anchors)
+    return ref.l1_drift_fused(pairs)
 
 
 class _BnTrain2dFn(torch.autograd.Function):

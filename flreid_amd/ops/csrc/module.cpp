@@ -54,6 +54,10 @@ extern "C" void flreid_bn_train_bwd(const void*, const void*, void*,
                                     const float*, const float*, const float*,
                                     float*, float*, float*, float*, int64_t,
                                     int, int, hipStream_t);
+extern "C" void flreid_drift_fwd(const int64_t*, const int*, float*, int,
+                                 hipStream_t);
+extern "C" void flreid_drift_bwd(const int64_t*, const int*, const float*,
+                                 float*, int, hipStream_t);
 }  // namespace flreid
 
 namespace py = pybind11;
@@ -190,6 +194,22 @@ PYBIND11_MODULE(_flreid_hip, m) {
                                       (float*)dgamma, (float*)dbeta,
                                       (float*)part_a, (float*)part_b, M, C,
                                       dtype, as_stream(stream));
+        });
+
+  m.def("drift_fwd",
+        [](uintptr_t ptrs, uintptr_t chunks, uintptr_t partials, int n_chunks,
+           uintptr_t stream) {
+          flreid::flreid_drift_fwd((const int64_t*)ptrs, (const int*)chunks,
+                                   (float*)partials, n_chunks,
+                                   as_stream(stream));
+        });
+
+  m.def("drift_bwd",
+        [](uintptr_t ptrs, uintptr_t chunks, uintptr_t gscale,
+           uintptr_t flat_grad, int n_chunks, uintptr_t stream) {
+          flreid::flreid_drift_bwd((const int64_t*)ptrs, (const int*)chunks,
+                                   (const float*)gscale, (float*)flat_grad,
+                                   n_chunks, as_stream(stream));
         });
 
   m.def("triplet_bwd",

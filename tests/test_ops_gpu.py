@@ -388,3 +388,28 @@ def test_bn_train_fused_1d_matches_torch():
     assert torch.allclose(x.grad, x_ref.grad, atol=1e-3, rtol=1e-2)
     assert torch.allclose(bn.weight.grad, bn_ref.weight.grad, atol=1e-3,
                           rtol=1e-2)
+
+
+def test_l1_drift_fused_matches_foreach():
+    """Multi-tensor HIP drift (one-pass fwd, flat sign bwd) vs the _foreach
+    reference (ops/csrc/drift.hip)."""
+    torch.manual_seed(2)
+    shapes = [(512, 512, 3, 3), (2048,), (8000, 2048), (7,)]
+    params = [torch.randn(s, device="cuda", requires_grad=True)
+              for s in shapes]
+    anchors = [torch.randn(s, device="cuda") for s in shapes]
+    pairs = [(p, a) for p, a in zip(params, anchors)]
+
+    loss = ops.l1_drift(pairs)
+    p2 = [p.detach().clone().requires_grad_(True) for p in params]
+    ref_loss = ref.l1_drift_fused([(q, a) for q, a in zip(p2, anchors)])
+    assert torch.allclose(loss, ref_loss, rtol=1e-6)
+
+    loss.backward()
+    ref_loss.backward()
+    for p, q in zip(params, p2):
+        assert torch.equal(p.grad, q.grad)
+
+    # cached-metadata path: second call must reuse tables and stay correct
+    loss2 = ops.l1_drift(pairs)
+    assert torch.allclose(loss2, ref_loss, rtol=1e-6)
