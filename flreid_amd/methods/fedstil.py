@@ -378,7 +378,10 @@ class Operator(BaseReIDOperator):
                         tap = model.eval_graphed(
                             "tap_fwd", lambda t: model.tap_forward(t)[1], d,
                             full=d.shape[0] == chunk)
-                    taps.append(tap.float())
+                    # keep the autocast dtype: the tap was COMPUTED in bf16,
+                    # so fp32 storage only doubles HBM traffic and adds a
+                    # cast per training step
+                    taps.append(tap)
                 del data_all, datas
             else:
                 taps, pids, classes = [], [], []
@@ -406,7 +409,8 @@ class Operator(BaseReIDOperator):
 
         loader = TensorBatches(all_data, all_pids, all_classes,
                                source_loader.batch_size, shuffle=True)
-        task_token = taps.reshape(taps.shape[0], -1).mean(dim=0).cpu()
+        task_token = taps.reshape(taps.shape[0], -1).mean(
+            dim=0, dtype=torch.float32).cpu()
         return loader, task_token
 
     def _train_step(self, model: Model, data: torch.Tensor,
