@@ -156,11 +156,13 @@ class Model(ModelModule):
             classes = proto_loader.classes
             feats = []
             chunk = 512
+            present = getattr(proto_loader, "present", lambda t: t)
             for i in range(0, protos.shape[0], chunk):
-                data = protos[i:i + chunk]
+                data = present(protos[i:i + chunk])
                 with autocast(device):
                     sf = self.eval_graphed("herd_fwd", self.head_forward, data,
                                            full=data.shape[0] == chunk)
+                # .float() also COPIES the graph's static output buffer
                 feats.append((sf[1] if isinstance(sf, tuple) else sf).float())
             feats = torch.cat(feats)
         else:
@@ -325,13 +327,27 @@ class TensorBatches:
     """
 
     def __init__(self, data: torch.Tensor, pids: torch.Tensor,
-                 classes: torch.Tensor, batch_size: int, shuffle: bool = True):
+                 classes: torch.Tensor, batch_size: int, shuffle: bool = True,
+                 nhwc_stored: bool = False):
+        # nhwc_stored: `data` rows are physically [H, W, C] (the tap's own
+        # channels-last layout, stored without transposition) and every
+        # yielded batch is a FREE channels-last [B, C, H, W] permute view —
+        # no per-step layout conversion in front of the head's convs
         self.data, self.pids, self.classes = data, pids, classes
         self.batch_size = batch_size
         self.shuffle = shuffle
+        self.nhwc_stored = nhwc_stored
+        # targets pre-staged on device: no per-step H2D in the train loop
+        self._pids_dev = pids.to(data.device) if data.is_cuda else None
         n = data.shape[0]
         self.drop_last = n % batch_size == 1
         self._len = n // batch_size + (0 if (self.drop_last or n % batch_size == 0) else 1)
+
+    def present(self, rows: torch.Tensor) -> torch.Tensor:
+        """Rows of `data` -> model-facing layout."""
+        if self.nhwc_stored and rows.dim() == 4:
+            return rows.permute(0, 3, 1, 2)
+        return rows
 
     def __len__(self):
         return self._len
@@ -344,8 +360,11 @@ class TensorBatches:
         for i in range(0, stop, self.batch_size):
             idx_cpu = order_cpu[i:i + self.batch_size]
             idx_dev = order_dev[i:i + self.batch_size]
-            yield (self.data.index_select(0, idx_dev),
-                   self.pids.index_select(0, idx_cpu),
+            pid_sel = (self._pids_dev.index_select(0, idx_dev)
+                       if self._pids_dev is not None
+                       else self.pids.index_select(0, idx_cpu))
+            yield (self.present(self.data.index_select(0, idx_dev)),
+                   pid_sel,
                    self.classes.index_select(0, idx_cpu))
 
 
@@ -378,10 +397,14 @@ class Operator(BaseReIDOperator):
                         tap = model.eval_graphed(
                             "tap_fwd", lambda t: model.tap_forward(t)[1], d,
                             full=d.shape[0] == chunk)
-                    # keep the autocast dtype: the tap was COMPUTED in bf16,
-                    # so fp32 storage only doubles HBM traffic and adds a
-                    # cast per training step
-                    taps.append(tap)
+                    # keep the autocast dtype (fp32 storage would only double
+                    # HBM traffic — the tap was COMPUTED in bf16) and store
+                    # rows physically [H, W, C]: the tap is channels-last, so
+                    # the permute is the identity on memory and every
+                    # rehearsal batch becomes a free channels-last view.
+                    # clone() is REQUIRED: eval_graphed returns the graph's
+                    # static output buffer, overwritten by the next replay
+                    taps.append(tap.permute(0, 2, 3, 1).clone())
                 del data_all, datas
             else:
                 taps, pids, classes = [], [], []
@@ -408,7 +431,8 @@ class Operator(BaseReIDOperator):
             all_data, all_pids, all_classes = taps, pids, classes
 
         loader = TensorBatches(all_data, all_pids, all_classes,
-                               source_loader.batch_size, shuffle=True)
+                               source_loader.batch_size, shuffle=True,
+                               nhwc_stored=on_gpu and all_data.dim() == 4)
         task_token = taps.reshape(taps.shape[0], -1).mean(
             dim=0, dtype=torch.float32).cpu()
         return loader, task_token
