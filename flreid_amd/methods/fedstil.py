@@ -475,8 +475,25 @@ class Operator(BaseReIDOperator):
             self._train_graphs = (key, gs)
         return self._train_graphs[1]
 
+    def _epoch_graph(self, model: Model, loader) -> "EpochGraph":
+        from flreid_amd.runtime.hipgraph import EpochGraph
+
+        n, batch = loader.data.shape[0], loader.batch_size
+        steps = n // batch
+        lr = float(self.optimizer.param_groups[0]["lr"])
+        shapes = tuple(tuple(p.shape) for p in
+                       self.optimizer.param_groups[0]["params"])
+        key = (tuple(loader.data.shape), batch, steps, lr, shapes)
+        cache = getattr(self, "_epoch_graph_cache", None)
+        if cache is None or cache[0] != key:
+            eg = EpochGraph(lambda d, t: self._train_step(model, d, t),
+                            steps, batch, loader.data, loader._pids_dev,
+                            loader.present)
+            self._epoch_graph_cache = (key, eg)   # latest only; old graph freed
+        return self._epoch_graph_cache[1]
+
     def invoke_train(self, model: Model, dataloader: DataLoader, **kwargs) -> Any:
-        from flreid_amd.runtime.hipgraph import hipgraph_enabled
+        from flreid_amd.runtime.hipgraph import epoch_graph_enabled, hipgraph_enabled
 
         train_acc = train_loss = 0.0
         batch_cnt = data_cnt = 0
@@ -490,29 +507,61 @@ class Operator(BaseReIDOperator):
         self.set_optimizer_parameters(model, capturable=use_graph)
         batch_size = getattr(proto_loader, "batch_size", None)
         _phase_head = phase("head_epoch"); _phase_head.__enter__()
-        gs = self._graphed_step(model, batch_size) if use_graph else None
-        warmups = 0
 
-        # device-side metric accumulators; always out-of-place adds so replay
-        # outputs (static tensors) are consumed before the next replay
-        acc_dev = loss_dev = None
-        for data, person_id, _class_id in proto_loader:
-            data = data.to(device, non_blocking=True)
-            target = person_id.to(device, non_blocking=True)
-            if gs is not None and data.shape[0] == batch_size:
-                if gs.ready:
-                    b_loss, b_acc = gs(data, target)
-                elif warmups < 2:
-                    b_loss, b_acc = gs.warmup(data, target)
-                    warmups += 1
+        use_epoch = (use_graph and epoch_graph_enabled()
+                     and isinstance(proto_loader, TensorBatches)
+                     and proto_loader.data.is_cuda
+                     and proto_loader._pids_dev is not None
+                     and proto_loader.data.shape[0] >= batch_size)
+        if use_epoch:
+            # the whole rehearsal epoch as one graph replay: S in-graph
+            # steps + an eager tail, preserving TensorBatches' shuffle and
+            # drop_last semantics (one host sync per epoch for the metrics)
+            n = proto_loader.data.shape[0]
+            steps = n // batch_size
+            order = torch.randperm(n)
+            eg = self._epoch_graph(model, proto_loader)
+            idx_flat = order[:steps * batch_size].to(device, non_blocking=True)
+            loss_dev, acc_dev = eg.run(proto_loader.data,
+                                       proto_loader._pids_dev, idx_flat)
+            batch_cnt, data_cnt = steps, steps * batch_size
+            stop = (steps * batch_size
+                    if (proto_loader.drop_last or n % batch_size == 0) else n)
+            if stop > steps * batch_size:
+                tail = order[steps * batch_size:stop].to(device)
+                d = proto_loader.present(
+                    proto_loader.data.index_select(0, tail))
+                t = proto_loader._pids_dev.index_select(0, tail)
+                b_loss, b_acc = self._train_step(model, d, t)
+                loss_dev = loss_dev + b_loss
+                acc_dev = acc_dev + b_acc
+                batch_cnt += 1
+                data_cnt += tail.numel()
+        else:
+            gs = self._graphed_step(model, batch_size) if use_graph else None
+            warmups = 0
+
+            # device-side metric accumulators; always out-of-place adds so
+            # replay outputs (static tensors) are consumed before the next
+            # replay
+            acc_dev = loss_dev = None
+            for data, person_id, _class_id in proto_loader:
+                data = data.to(device, non_blocking=True)
+                target = person_id.to(device, non_blocking=True)
+                if gs is not None and data.shape[0] == batch_size:
+                    if gs.ready:
+                        b_loss, b_acc = gs(data, target)
+                    elif warmups < 2:
+                        b_loss, b_acc = gs.warmup(data, target)
+                        warmups += 1
+                    else:
+                        b_loss, b_acc = gs.capture(data, target)
                 else:
-                    b_loss, b_acc = gs.capture(data, target)
-            else:
-                b_loss, b_acc = self._train_step(model, data, target)
-            acc_dev = b_acc.clone() if acc_dev is None else acc_dev + b_acc
-            loss_dev = b_loss.clone() if loss_dev is None else loss_dev + b_loss
-            data_cnt += len(data)
-            batch_cnt += 1
+                    b_loss, b_acc = self._train_step(model, data, target)
+                acc_dev = b_acc.clone() if acc_dev is None else acc_dev + b_acc
+                loss_dev = b_loss.clone() if loss_dev is None else loss_dev + b_loss
+                data_cnt += len(data)
+                batch_cnt += 1
         _phase_head.__exit__(None, None, None)
         if acc_dev is not None:       # single host sync per epoch
             train_acc = float(acc_dev)

@@ -75,6 +75,81 @@ class GraphedStep:
         return self.static_outputs
 
 
+def epoch_graph_enabled() -> bool:
+    """Whole-epoch capture is OPT-IN (FLREID_EPOCH_GRAPH=1): measured
+    steady-state −4 ms/round over the per-step graphs, but instantiating the
+    ~8000-node graph costs ~1 s once — worth it for production runs of many
+    hundreds of rounds, a loss for short benchmarks."""
+    return (hipgraph_enabled()
+            and os.environ.get("FLREID_EPOCH_GRAPH", "0") == "1")
+
+
+class EpochGraph:
+    """Captures an ENTIRE rehearsal epoch — S full training steps over a
+    device-resident sample store — as ONE hipGraph.
+
+    Per-epoch inputs are data, not topology: the store content, the target
+    ids, and the [S·B] shuffled batch-index matrix are copied into static
+    buffers, then one replay executes all S steps (index_select → forward →
+    backward → Adam, ~200 kernels each) back-to-back with zero host
+    involvement between steps.  The per-step GraphedStep already removed the
+    CPU launch cost; this removes the remaining per-step replay/submission
+    gaps and the out-of-graph batch gathers.
+
+    Same train-exactly-once protocol as GraphedStep: two warmup epochs run
+    eagerly, the third records the graph and immediately replays it, later
+    epochs replay.  Metric outputs are two device scalars (Σloss, Σacc)
+    accumulated inside the graph — one host sync per epoch at most.
+    """
+
+    def __init__(self, step_fn: Callable, steps: int, batch: int,
+                 store_template: torch.Tensor, pids_template: torch.Tensor,
+                 present: Callable):
+        self.step_fn = step_fn
+        self.S, self.B = steps, batch
+        self.store = torch.empty_like(store_template)
+        self.pids = torch.empty_like(pids_template)
+        self.idx = torch.empty(steps * batch, dtype=torch.long,
+                               device=store_template.device)
+        self.present = present
+        self.graph = None
+        self.static_out = None
+        self._warmups = 0
+        self._stream = torch.cuda.Stream()
+
+    def _epoch_body(self):
+        loss_t = acc_t = None
+        idx2 = self.idx.view(self.S, self.B)
+        for s in range(self.S):
+            data = self.present(self.store.index_select(0, idx2[s]))
+            target = self.pids.index_select(0, idx2[s])
+            loss, acc = self.step_fn(data, target)
+            loss_t = loss.clone() if loss_t is None else loss_t + loss
+            acc_t = acc.clone() if acc_t is None else acc_t + acc
+        return loss_t, acc_t
+
+    def run(self, store: torch.Tensor, pids: torch.Tensor,
+            idx_flat: torch.Tensor):
+        self.store.copy_(store, non_blocking=True)
+        self.pids.copy_(pids, non_blocking=True)
+        self.idx.copy_(idx_flat, non_blocking=True)
+        if self.graph is not None:
+            self.graph.replay()
+            return self.static_out
+        if self._warmups < 1:
+            self._warmups += 1
+            self._stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._stream):
+                out = self._epoch_body()
+            torch.cuda.current_stream().wait_stream(self._stream)
+            return out
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.static_out = self._epoch_body()
+        self.graph.replay()
+        return self.static_out
+
+
 # ---------------------------------------------------------------------------
 # lightweight phase timing (FLREID_PHASE_TIMERS=1): per-phase wall clock with
 # a device sync at each boundary — for finding where a round's time goes

@@ -81,3 +81,31 @@ def test_gpu_round(method, tmp_path, monkeypatch):
 def test_smoke_entry():
     import __graft_entry__
     __graft_entry__.smoke()
+
+
+def test_gpu_round_epoch_graph(tmp_path, monkeypatch):
+    """FedSTIL with whole-epoch hipGraph capture (FLREID_EPOCH_GRAPH=1,
+    runtime/hipgraph.py::EpochGraph): rounds must train and validate like
+    the per-step-graph path."""
+    from flreid_amd import ops
+    assert ops.extension_available()
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.setenv("FLREID_EPOCH_GRAPH", "1")
+    from flreid_amd.parallel.comm import FedContext
+    from flreid_amd.runtime.experiment import ExperimentStage
+
+    common = _common(tmp_path)
+    cfg = _exp("fedstil")
+    cfg["exp_name"] = "gpu-fedstil-epochgraph"
+    cfg["exp_opts"]["comm_rounds"] = 4   # reach capture + replay rounds
+    cfg["exp_opts"]["val_interval"] = 4
+    ctx = FedContext(device="cuda:0")
+    stage = ExperimentStage(common, [cfg], ctx=ctx)
+    log = stage.run_experiment(cfg)
+    data = log.records["data"]
+    r4 = data["client-0"].get("4", {})
+    assert any("val_map" in v for v in r4.values())
+    for r in ("1", "2", "3", "4"):
+        rec = data["client-0"].get(r, {})
+        tr = [v for v in rec.values() if "tr_loss" in v]
+        assert tr and all(v["tr_loss"] == v["tr_loss"] for v in tr)  # finite
