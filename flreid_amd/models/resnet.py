@@ -21,6 +21,7 @@ from typing import List, Type, Union
 
 import torch
 import torch.nn as nn
+import torch.nn.functional as F
 
 from flreid_amd import ops
 from flreid_amd.tools.logger import Logger
@@ -37,18 +38,24 @@ class EvalFusedBatchNorm2d(nn.BatchNorm2d):
         cached prototype features): single-kernel fwd/bwd with in-kernel
         statistics and running-stat update, replacing MIOpen's 5-kernel
         chain + fp32 autocast round-trip per layer (see ops/csrc/bn_train.hip).
-    CPU and out-of-regime shapes fall through to torch."""
+    When `fuse_relu` is set (bn1/bn2 inside the residual blocks and the stem
+    BN — everywhere the model applies BN→ReLU directly) the ReLU runs inside
+    the same kernels; the torch fallback applies it explicitly so CPU
+    semantics are identical.  CPU and out-of-regime shapes fall through."""
+
+    fuse_relu = False
 
     def forward(self, x):
         if not self.training:
-            out = ops.bn_eval_2d(x, self)
+            out = ops.bn_eval_2d(x, self, relu=self.fuse_relu)
             if out is not None:
                 return out
         else:
-            out = ops.bn_train_2d(x, self)
+            out = ops.bn_train_2d(x, self, relu=self.fuse_relu)
             if out is not None:
                 return out
-        return super().forward(x)
+        out = super().forward(x)
+        return F.relu(out, inplace=True) if self.fuse_relu else out
 
 
 class EvalFusedBatchNorm1d(nn.BatchNorm1d):
@@ -84,6 +91,7 @@ class BasicBlock(nn.Module):
         super().__init__()
         self.conv1 = conv3x3(cin, planes, stride)
         self.bn1 = EvalFusedBatchNorm2d(planes)
+        self.bn1.fuse_relu = True
         self.relu = nn.ReLU(inplace=True)
         self.conv2 = conv3x3(planes, planes)
         self.bn2 = EvalFusedBatchNorm2d(planes)
@@ -91,7 +99,7 @@ class BasicBlock(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))          # ReLU fused into bn1
         out = self.bn2(self.conv2(out))
         return self.relu(out + identity)
 
@@ -104,8 +112,10 @@ class Bottleneck(nn.Module):
         super().__init__()
         self.conv1 = conv1x1(cin, planes)
         self.bn1 = EvalFusedBatchNorm2d(planes)
+        self.bn1.fuse_relu = True
         self.conv2 = conv3x3(planes, planes, stride)
         self.bn2 = EvalFusedBatchNorm2d(planes)
+        self.bn2.fuse_relu = True
         self.conv3 = conv1x1(planes, planes * self.expansion)
         self.bn3 = EvalFusedBatchNorm2d(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
@@ -113,8 +123,8 @@ class Bottleneck(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn1(self.conv1(x))          # ReLU fused into bn1/bn2
+        out = self.bn2(self.conv2(out))
         out = self.bn3(self.conv3(out))
         return self.relu(out + identity)
 
@@ -128,6 +138,7 @@ class ResNetTrunk(nn.Module):
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = EvalFusedBatchNorm2d(64)
+        self.bn1.fuse_relu = True
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_stage(block, 64, layers[0])
@@ -158,7 +169,7 @@ class ResNetTrunk(nn.Module):
                 nn.init.constant_(m.bias, 0.0)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x)
         return torch.flatten(x, 1)
@@ -245,7 +256,7 @@ class ResNetReID(nn.Module):
         tap_value = None
         b = self.base
         stages = [
-            lambda t: b.maxpool(b.relu(b.bn1(b.conv1(t)))),
+            lambda t: b.maxpool(b.bn1(b.conv1(t))),
             b.layer1, b.layer2, b.layer3, b.layer4,
         ]
         for idx in range(start, 5):

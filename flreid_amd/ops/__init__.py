@@ -414,7 +414,7 @@ def adaptive_linear(x: torch.Tensor, gw: torch.Tensor, atten: torch.Tensor,
     return torch.nn.functional.linear(x, theta, bias)
 
 
-def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
+def bn_eval_2d(x: torch.Tensor, bn, relu: bool = False) -> Optional[torch.Tensor]:
     """Fused eval-mode BatchNorm2d (one coalesced pass; MIOpen's inference
     kernel measured ~0.3 TB/s on ReID shapes).  Returns None when the fused
     path does not apply (train mode, CPU, missing stats/extension)."""
@@ -441,8 +441,8 @@ def bn_eval_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
                 bn.bias.detach().float().contiguous().data_ptr(),
                 bn.running_mean.float().contiguous().data_ptr(),
                 bn.running_var.float().contiguous().data_ptr(),
-                x.numel(), c, hw, float(bn.eps), int(nhwc), _dt(x),
-                _stream())
+                x.numel(), c, hw, float(bn.eps), int(nhwc), int(relu),
+                _dt(x), _stream())
     return out
 
 
@@ -534,7 +534,7 @@ class _BnTrain2dFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
-                nbt):
+                nbt, relu):
         ext = _ext_or_raise("bn_train")
         c = x.shape[1]
         m = x.numel() // c
@@ -556,13 +556,14 @@ class _BnTrain2dFn(torch.autograd.Function):
             part[0].data_ptr(), part[1].data_ptr(),
             nbt.data_ptr() if nbt is not None else 0, m, c,
             float(momentum), float(eps), float(m) / float(m - 1),
-            _dt(x), _stream())
-        ctx.save_for_backward(x, gamma, smean, sinv)
+            int(relu), _dt(x), _stream())
+        ctx.save_for_backward(x, gamma, beta, smean, sinv)
+        ctx.relu = bool(relu)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, gamma, smean, sinv = ctx.saved_tensors
+        x, gamma, beta, smean, sinv = ctx.saved_tensors
         if x.dim() == 4:
             if not dy.is_contiguous(memory_format=torch.channels_last):
                 dy = dy.contiguous(memory_format=torch.channels_last)
@@ -580,14 +581,16 @@ class _BnTrain2dFn(torch.autograd.Function):
         part = torch.empty(2, nslab, c, device=x.device, dtype=torch.float32)
         ext.bn_train_bwd(x.data_ptr(), dy.data_ptr(), dx.data_ptr(),
                          gamma.detach().contiguous().data_ptr(),
+                         beta.detach().contiguous().data_ptr(),
                          smean.data_ptr(), sinv.data_ptr(),
                          dgamma.data_ptr(), dbeta.data_ptr(),
                          part[0].data_ptr(), part[1].data_ptr(),
-                         m, c, _dt(x), _stream())
-        return dx, dgamma, dbeta, None, None, None, None, None
+                         m, c, int(ctx.relu), _dt(x), _stream())
+        return dx, dgamma, dbeta, None, None, None, None, None, None
 
 
-def bn_train_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
+def bn_train_2d(x: torch.Tensor, bn,
+                relu: bool = False) -> Optional[torch.Tensor]:
     """Fused TRAINING-mode BatchNorm for the head-epoch regime.  Applies to
     CUDA channels-last 4-D tensors (BatchNorm2d) and contiguous 2-D tensors
     (BatchNorm1d — the BNNeck bottleneck) with C % 64 == 0 and a small row
@@ -623,7 +626,7 @@ def bn_train_2d(x: torch.Tensor, bn) -> Optional[torch.Tensor]:
     else:
         rm = rv = None
     return _BnTrain2dFn.apply(x, bn.weight, bn.bias, rm, rv,
-                              bn.momentum, bn.eps, nbt)
+                              bn.momentum, bn.eps, nbt, relu)
 
 
 def conv3x3_fwd_nhwc(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:

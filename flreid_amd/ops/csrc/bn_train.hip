@@ -82,8 +82,9 @@ template <typename T, bool IS_BWD>
 __global__ __launch_bounds__(256) void bn_partial_kernel(
     const T* __restrict__ X, const T* __restrict__ DY,
     const float* __restrict__ smean, const float* __restrict__ sinv,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ part_a, float* __restrict__ part_b, int64_t M,
-    int C) {
+    int C, int relu) {
   const int c0 = blockIdx.x * BN_CG;
   const int q = threadIdx.x & 15;       // channel quad within the group
   const int r = threadIdx.x >> 4;       // stripe within the block
@@ -94,12 +95,16 @@ __global__ __launch_bounds__(256) void bn_partial_kernel(
   __shared__ float la[BN_STRIPES][BN_CG + 4];
   __shared__ float lb[BN_STRIPES][BN_CG + 4];
 
-  float mn[4], iv[4];
+  float mn[4], iv[4], sc[4], sh[4];
   if (IS_BWD) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       mn[j] = smean[c + j];
       iv[j] = sinv[c + j];
+      if (relu) {                    // recompute pre-relu output sign
+        sc[j] = gamma[c + j] * iv[j];
+        sh[j] = beta[c + j] - mn[j] * sc[j];
+      }
     }
   }
 
@@ -113,6 +118,7 @@ __global__ __launch_bounds__(256) void bn_partial_kernel(
       bn_load4(DY + row * C + c, dy);
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
+        if (relu && x[j] * sc[j] + sh[j] <= 0.f) continue;  // dy masked by y>0
         sa[j] += dy[j];
         sb[j] += dy[j] * (x[j] - mn[j]) * iv[j];
       }
@@ -152,7 +158,7 @@ __global__ __launch_bounds__(256) void bn_finalize_fwd_kernel(
     float* __restrict__ smean, float* __restrict__ sinv,
     const float* __restrict__ part_a, const float* __restrict__ part_b,
     unsigned long long* __restrict__ nbt, int64_t M, int C, float momentum,
-    float eps, float unbiased) {
+    float eps, float unbiased, int relu) {
   const int c0 = blockIdx.x * BN_CG;
   const int q = threadIdx.x & 15;
   const int r = threadIdx.x >> 4;
@@ -202,7 +208,10 @@ __global__ __launch_bounds__(256) void bn_finalize_fwd_kernel(
     float v[4];
     bn_load4(X + row * C + c, v);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) v[j] = v[j] * scale[j] + shift[j];
+    for (int j = 0; j < 4; ++j) {
+      v[j] = v[j] * scale[j] + shift[j];
+      if (relu) v[j] = fmaxf(v[j], 0.f);
+    }
     bn_store4(Y + row * C + c, v);
   }
 }
@@ -212,10 +221,11 @@ __global__ __launch_bounds__(256) void bn_finalize_fwd_kernel(
 template <typename T>
 __global__ __launch_bounds__(256) void bn_finalize_bwd_kernel(
     const T* __restrict__ X, const T* __restrict__ DY, T* __restrict__ DX,
-    const float* __restrict__ gamma, const float* __restrict__ smean,
-    const float* __restrict__ sinv, float* __restrict__ dgamma,
-    float* __restrict__ dbeta, const float* __restrict__ part_a,
-    const float* __restrict__ part_b, int64_t M, int C) {
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ smean, const float* __restrict__ sinv,
+    float* __restrict__ dgamma, float* __restrict__ dbeta,
+    const float* __restrict__ part_a, const float* __restrict__ part_b,
+    int64_t M, int C, int relu) {
   const int c0 = blockIdx.x * BN_CG;
   const int q = threadIdx.x & 15;
   const int r = threadIdx.x >> 4;
@@ -242,7 +252,7 @@ __global__ __launch_bounds__(256) void bn_finalize_bwd_kernel(
   }
   __syncthreads();
 
-  float g_iv[4], mn[4], iv[4], mdy[4], mdyx[4];
+  float g_iv[4], mn[4], iv[4], mdy[4], mdyx[4], sh[4];
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
     mn[j] = smean[c + j];
@@ -250,6 +260,7 @@ __global__ __launch_bounds__(256) void bn_finalize_bwd_kernel(
     g_iv[j] = gamma[c + j] * iv[j];
     mdy[j] = lmdy[q * 4 + j];
     mdyx[j] = lmdyx[q * 4 + j];
+    if (relu) sh[j] = beta[c + j] - mn[j] * g_iv[j];
   }
   for (int64_t row = g; row < M; row += S) {
     float x[4], dy[4], dx[4];
@@ -258,7 +269,9 @@ __global__ __launch_bounds__(256) void bn_finalize_bwd_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const float xhat = (x[j] - mn[j]) * iv[j];
-      dx[j] = g_iv[j] * (dy[j] - mdy[j] - xhat * mdyx[j]);
+      float d = dy[j];
+      if (relu && x[j] * g_iv[j] + sh[j] <= 0.f) d = 0.f;
+      dx[j] = g_iv[j] * (d - mdy[j] - xhat * mdyx[j]);
     }
     bn_store4(DX + row * C + c, dx);
   }
@@ -287,53 +300,57 @@ extern "C" void flreid_bn_train_fwd(const void* X, void* Y, const float* gamma,
                                     float* part_a, float* part_b,
                                     unsigned long long* nbt, int64_t M, int C,
                                     float momentum, float eps, float unbiased,
-                                    int dtype, hipStream_t stream) {
+                                    int relu, int dtype, hipStream_t stream) {
   if (C % BN_CG != 0) throw std::runtime_error("bn_train: C % 64 != 0");
   dim3 grid(C / BN_CG, bn_nslab(M, C)), block(256);
   if (dtype == kBF16) {
     hipLaunchKernelGGL((bn_partial_kernel<__hip_bfloat16, false>), grid,
                        block, 0, stream, (const __hip_bfloat16*)X, nullptr,
-                       nullptr, nullptr, part_a, part_b, M, C);
+                       nullptr, nullptr, nullptr, nullptr, part_a, part_b, M,
+                       C, 0);
     hipLaunchKernelGGL(bn_finalize_fwd_kernel<__hip_bfloat16>, grid, block, 0,
                        stream, (const __hip_bfloat16*)X, (__hip_bfloat16*)Y,
                        gamma, beta, rmean, rvar, smean, sinv, part_a, part_b,
-                       nbt, M, C, momentum, eps, unbiased);
+                       nbt, M, C, momentum, eps, unbiased, relu);
   } else {
     hipLaunchKernelGGL((bn_partial_kernel<float, false>), grid, block, 0,
                        stream, (const float*)X, nullptr, nullptr, nullptr,
-                       part_a, part_b, M, C);
+                       nullptr, nullptr, part_a, part_b, M, C, 0);
     hipLaunchKernelGGL(bn_finalize_fwd_kernel<float>, grid, block, 0, stream,
                        (const float*)X, (float*)Y, gamma, beta, rmean, rvar,
                        smean, sinv, part_a, part_b, nbt, M, C, momentum, eps,
-                       unbiased);
+                       unbiased, relu);
   }
   HIP_CHECK(hipGetLastError());
 }
 
 extern "C" void flreid_bn_train_bwd(const void* X, const void* DY, void* DX,
-                                    const float* gamma, const float* smean,
-                                    const float* sinv, float* dgamma,
-                                    float* dbeta, float* part_a,
-                                    float* part_b, int64_t M, int C,
-                                    int dtype, hipStream_t stream) {
+                                    const float* gamma, const float* beta,
+                                    const float* smean, const float* sinv,
+                                    float* dgamma, float* dbeta,
+                                    float* part_a, float* part_b, int64_t M,
+                                    int C, int relu, int dtype,
+                                    hipStream_t stream) {
   if (C % BN_CG != 0) throw std::runtime_error("bn_train: C % 64 != 0");
   dim3 grid(C / BN_CG, bn_nslab(M, C)), block(256);
   if (dtype == kBF16) {
     hipLaunchKernelGGL((bn_partial_kernel<__hip_bfloat16, true>), grid, block,
                        0, stream, (const __hip_bfloat16*)X,
-                       (const __hip_bfloat16*)DY, smean, sinv, part_a, part_b,
-                       M, C);
+                       (const __hip_bfloat16*)DY, smean, sinv, gamma, beta,
+                       part_a, part_b, M, C, relu);
     hipLaunchKernelGGL(bn_finalize_bwd_kernel<__hip_bfloat16>, grid, block, 0,
                        stream, (const __hip_bfloat16*)X,
                        (const __hip_bfloat16*)DY, (__hip_bfloat16*)DX, gamma,
-                       smean, sinv, dgamma, dbeta, part_a, part_b, M, C);
+                       beta, smean, sinv, dgamma, dbeta, part_a, part_b, M, C,
+                       relu);
   } else {
     hipLaunchKernelGGL((bn_partial_kernel<float, true>), grid, block, 0,
                        stream, (const float*)X, (const float*)DY, smean, sinv,
-                       part_a, part_b, M, C);
+                       gamma, beta, part_a, part_b, M, C, relu);
     hipLaunchKernelGGL(bn_finalize_bwd_kernel<float>, grid, block, 0, stream,
                        (const float*)X, (const float*)DY, (float*)DX, gamma,
-                       smean, sinv, dgamma, dbeta, part_a, part_b, M, C);
+                       beta, smean, sinv, dgamma, dbeta, part_a, part_b, M, C,
+                       relu);
   }
   HIP_CHECK(hipGetLastError());
 }

@@ -217,42 +217,43 @@ __global__ void bn_eval_kernel(const T* __restrict__ x, T* __restrict__ y,
                                const float* __restrict__ beta,
                                const float* __restrict__ mean,
                                const float* __restrict__ var, int64_t numel,
-                               int C, int64_t HW, float eps) {
+                               int C, int64_t HW, float eps, int relu) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= numel) return;
   const int c = NHWC ? (int)(i % C) : (int)((i / HW) % C);
   const float inv = __frsqrt_rn(var[c] + eps);
-  const float v = (load_as_float(x, i) - mean[c]) * inv * gamma[c] + beta[c];
+  float v = (load_as_float(x, i) - mean[c]) * inv * gamma[c] + beta[c];
+  if (relu) v = fmaxf(v, 0.f);
   store_from_float(y, i, v);
 }
 
 extern "C" void flreid_bn_eval(const void* x, void* y, const float* gamma,
                                const float* beta, const float* mean,
                                const float* var, int64_t numel, int C,
-                               int64_t HW, float eps, int nhwc, int dtype,
-                               hipStream_t stream) {
+                               int64_t HW, float eps, int nhwc, int relu,
+                               int dtype, hipStream_t stream) {
   constexpr int BLOCK = 256;
   dim3 grid((unsigned)((numel + BLOCK - 1) / BLOCK)), block(BLOCK);
   if (dtype == kF32) {
     if (nhwc)
       hipLaunchKernelGGL((bn_eval_kernel<float, true>), grid, block, 0, stream,
                          (const float*)x, (float*)y, gamma, beta, mean, var,
-                         numel, C, HW, eps);
+                         numel, C, HW, eps, relu);
     else
       hipLaunchKernelGGL((bn_eval_kernel<float, false>), grid, block, 0,
                          stream, (const float*)x, (float*)y, gamma, beta, mean,
-                         var, numel, C, HW, eps);
+                         var, numel, C, HW, eps, relu);
   } else {
     if (nhwc)
       hipLaunchKernelGGL((bn_eval_kernel<__hip_bfloat16, true>), grid, block,
                          0, stream, (const __hip_bfloat16*)x,
                          (__hip_bfloat16*)y, gamma, beta, mean, var, numel, C,
-                         HW, eps);
+                         HW, eps, relu);
     else
       hipLaunchKernelGGL((bn_eval_kernel<__hip_bfloat16, false>), grid, block,
                          0, stream, (const __hip_bfloat16*)x,
                          (__hip_bfloat16*)y, gamma, beta, mean, var, numel, C,
-                         HW, eps);
+                         HW, eps, relu);
   }
   HIP_CHECK(hipGetLastError());
 }

@@ -41,7 +41,7 @@ extern "C" void flreid_adaptive_linear_fwd(const void*, const float*,
                                            int, int, hipStream_t);
 extern "C" void flreid_bn_eval(const void*, void*, const float*, const float*,
                                const float*, const float*, int64_t, int,
-                               int64_t, float, int, int, hipStream_t);
+                               int64_t, float, int, int, int, hipStream_t);
 extern "C" void flreid_conv3x3_fwd(const void*, const float*, void*, int, int,
                                    int, int, int, hipStream_t);
 extern "C" int flreid_bn_train_nslab(int64_t, int);
@@ -49,11 +49,12 @@ extern "C" void flreid_bn_train_fwd(const void*, void*, const float*,
                                     const float*, float*, float*, float*,
                                     float*, float*, float*,
                                     unsigned long long*, int64_t, int, float,
-                                    float, float, int, hipStream_t);
+                                    float, float, int, int, hipStream_t);
 extern "C" void flreid_bn_train_bwd(const void*, const void*, void*,
                                     const float*, const float*, const float*,
-                                    float*, float*, float*, float*, int64_t,
-                                    int, int, hipStream_t);
+                                    const float*, float*, float*, float*,
+                                    float*, int64_t, int, int, int,
+                                    hipStream_t);
 extern "C" void flreid_drift_fwd(const int64_t*, const int*, float*, int,
                                  hipStream_t);
 extern "C" void flreid_drift_bwd(const int64_t*, const int*, const float*,
@@ -148,11 +149,12 @@ PYBIND11_MODULE(_flreid_hip, m) {
   m.def("bn_eval",
         [](uintptr_t x, uintptr_t y, uintptr_t gamma, uintptr_t beta,
            uintptr_t mean, uintptr_t var, int64_t numel, int C, int64_t HW,
-           float eps, int nhwc, int dtype, uintptr_t stream) {
+           float eps, int nhwc, int relu, int dtype, uintptr_t stream) {
           flreid::flreid_bn_eval((const void*)x, (void*)y,
                                  (const float*)gamma, (const float*)beta,
                                  (const float*)mean, (const float*)var, numel,
-                                 C, HW, eps, nhwc, dtype, as_stream(stream));
+                                 C, HW, eps, nhwc, relu, dtype,
+                                 as_stream(stream));
         });
 
   m.def("conv3x3_fwd",
@@ -171,29 +173,30 @@ PYBIND11_MODULE(_flreid_hip, m) {
         [](uintptr_t x, uintptr_t y, uintptr_t gamma, uintptr_t beta,
            uintptr_t rmean, uintptr_t rvar, uintptr_t smean, uintptr_t sinv,
            uintptr_t part_a, uintptr_t part_b, uintptr_t nbt, int64_t M,
-           int C, float momentum, float eps, float unbiased, int dtype,
-           uintptr_t stream) {
+           int C, float momentum, float eps, float unbiased, int relu,
+           int dtype, uintptr_t stream) {
           flreid::flreid_bn_train_fwd((const void*)x, (void*)y,
                                       (const float*)gamma, (const float*)beta,
                                       (float*)rmean, (float*)rvar,
                                       (float*)smean, (float*)sinv,
                                       (float*)part_a, (float*)part_b,
                                       (unsigned long long*)nbt, M, C,
-                                      momentum, eps, unbiased, dtype,
+                                      momentum, eps, unbiased, relu, dtype,
                                       as_stream(stream));
         });
 
   m.def("bn_train_bwd",
         [](uintptr_t x, uintptr_t dy, uintptr_t dx, uintptr_t gamma,
-           uintptr_t smean, uintptr_t sinv, uintptr_t dgamma, uintptr_t dbeta,
-           uintptr_t part_a, uintptr_t part_b, int64_t M, int C, int dtype,
-           uintptr_t stream) {
+           uintptr_t beta, uintptr_t smean, uintptr_t sinv, uintptr_t dgamma,
+           uintptr_t dbeta, uintptr_t part_a, uintptr_t part_b, int64_t M,
+           int C, int relu, int dtype, uintptr_t stream) {
           flreid::flreid_bn_train_bwd((const void*)x, (const void*)dy,
                                       (void*)dx, (const float*)gamma,
-                                      (const float*)smean, (const float*)sinv,
-                                      (float*)dgamma, (float*)dbeta,
-                                      (float*)part_a, (float*)part_b, M, C,
-                                      dtype, as_stream(stream));
+                                      (const float*)beta, (const float*)smean,
+                                      (const float*)sinv, (float*)dgamma,
+                                      (float*)dbeta, (float*)part_a,
+                                      (float*)part_b, M, C, relu, dtype,
+                                      as_stream(stream));
         });
 
   m.def("drift_fwd",

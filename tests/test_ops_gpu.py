@@ -347,6 +347,40 @@ def test_bn_train_fused_matches_torch():
                                   atol=atol * 10, rtol=5e-2)
 
 
+def test_bn_train_fused_relu_matches_torch():
+    """relu=True fuses BN→ReLU: fwd clamps, bwd masks dy by y>0
+    (bn1/bn2 + stem inside the residual blocks)."""
+    import copy
+    import torch.nn as nn
+    torch.manual_seed(3)
+    bn = nn.BatchNorm2d(128).cuda().train()
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.data.uniform_(-1, 1)
+    bn_ref = copy.deepcopy(bn)
+    x = (torch.randn(16, 128, 8, 4, device="cuda")
+         .to(memory_format=torch.channels_last).requires_grad_(True))
+    x_ref = x.detach().clone().requires_grad_(True)
+    y = ops.bn_train_2d(x, bn, relu=True)
+    assert y is not None and (y >= 0).all()
+    y_ref = torch.relu(bn_ref(x_ref))
+    assert torch.allclose(y, y_ref, atol=1e-4, rtol=1e-3)
+    dy = torch.randn_like(y_ref)
+    y.backward(dy)
+    y_ref.backward(dy)
+    assert torch.allclose(x.grad, x_ref.grad, atol=1e-3, rtol=1e-2)
+    assert torch.allclose(bn.weight.grad, bn_ref.weight.grad, atol=1e-3,
+                          rtol=1e-2)
+    assert torch.allclose(bn.bias.grad, bn_ref.bias.grad, atol=1e-3,
+                          rtol=1e-2)
+
+    # eval relu path
+    bn.eval()
+    xe = torch.randn(8, 128, 8, 4, device="cuda") \
+        .to(memory_format=torch.channels_last)
+    ye = ops.bn_eval_2d(xe, bn, relu=True)
+    assert torch.allclose(ye, torch.relu(bn(xe)), atol=1e-5, rtol=1e-5)
+
+
 def test_bn_train_fused_declines_out_of_regime():
     import torch.nn as nn
     bn = nn.BatchNorm2d(64).cuda().train()
