@@ -121,6 +121,29 @@ def main():
         results[f"bn_train_c{c}_fused"] = bench(fused_bn, n)
         results[f"bn_train_c{c}_miopen"] = bench(miopen_bn, n)
 
+    # fused multi-tensor L1 drift (FedSTIL per-step regulariser) vs _foreach
+    shapes = [(512, 512, 3, 3), (2048, 512, 1, 1), (512, 2048, 1, 1),
+              (8000, 2048)] * 3
+    params = [torch.randn(s, device="cuda", requires_grad=True)
+              for s in shapes]
+    anchors = [torch.randn(s, device="cuda") for s in shapes]
+    pairs = list(zip(params, anchors))
+
+    def fused_drift():
+        loss = ops.l1_drift(pairs)
+        loss.backward()
+        for p in params:
+            p.grad = None
+
+    def foreach_drift():
+        loss = ref.l1_drift_fused(pairs)
+        loss.backward()
+        for p in params:
+            p.grad = None
+
+    results["drift_fused"] = bench(fused_drift, n)
+    results["drift_foreach"] = bench(foreach_drift, n)
+
     for k in sorted(results):
         v = results[k]
         print(f"{k:28s} {v:10.3f}" + (" ms" if "_TF" not in k and "GBps" not in k else ""))
