@@ -240,7 +240,15 @@ class Server(ServerModule):
 
         self.client_aw = [s["incremental_aw"] for _c, s in states.items()]
         if len(self.client_aw) >= self.model.kb_cnt:
-            sampled = random.sample(self.client_aw, self.model.kb_cnt)
+            # REPLICA-DETERMINISTIC sampling: every rank hosts a server
+            # replica and each must build the SAME knowledge base, but the
+            # ranks' global RNG streams differ (they trained different
+            # clients) — a private generator seeded from the aggregation
+            # counter keeps replicas bit-identical and placement-invariant
+            self._kb_rounds = getattr(self, "_kb_rounds", 0) + 1
+            rng = random.Random(0xFED0 + self._kb_rounds * 1009
+                                + len(self.client_aw))
+            sampled = rng.sample(self.client_aw, self.model.kb_cnt)
             kb = {}
             for name in sampled[0]:
                 kb[name[:-len(".aw")] + ".aw_kb"] = torch.cat(

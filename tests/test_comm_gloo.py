@@ -302,3 +302,101 @@ def test_fedstil_distributed_equals_single(tmp_path):
     assert set(single) == set(dist_state)
     for n in single:
         assert torch.allclose(single[n], dist_state[n], atol=1e-6), n
+
+
+def _fedweit_cfg(tmpdir, tag):
+    common = {
+        "datasets_dir": "synthetic://ids=4,train=3,query=2,gallery=3,hw=32x16,idspace=48",
+        "checkpoints_dir": os.path.join(tmpdir, f"ckw-{tag}"),
+        "logs_dir": os.path.join(tmpdir, f"lgw-{tag}"),
+        "parallel": 1, "device": ["cpu"], "defaults": {},
+    }
+    exp = {
+        "exp_name": "dist-fedweit", "exp_method": "fedweit", "random_seed": 9,
+        "exp_opts": {"comm_rounds": 2, "val_interval": 0, "online_clients": 3,
+                     "initial_validation": False, "persist_comm_ckpts": False},
+        "model_opts": {"name": "resnet18", "num_classes": 64, "last_stride": 1,
+                       "neck": "bnneck", "lambda_l1": 5e-6, "lambda_l2": 1e-3,
+                       "lambda_mask": 0.0, "kb_cnt": 2,
+                       "fine_tuning": ["classifier"]},
+        "criterion_opts": {"name": "cross_entropy", "num_classes": 64,
+                           "epsilon": 0.1},
+        "optimizer_opts": {"name": "adam", "lr": 1e-3, "weight_decay": 1e-5},
+        "scheduler_opts": {"name": "step_lr", "step_size": 5},
+        "task_opts": {"sustain_rounds": 2, "train_epochs": 1,
+                      "augment_opts": {"level": "none", "img_size": [32, 16],
+                                       "norm_mean": [0.485, 0.456, 0.406],
+                                       "norm_std": [0.229, 0.224, 0.225]},
+                      "loader_opts": {"batch_size": 8, "num_workers": 0,
+                                      "pin_memory": False,
+                                      "persistent_workers": False,
+                                      "multiprocessing_context": None}},
+        "server": {"server_name": "server"},
+        "clients": [{"client_name": f"client-{i}", "tasks": [f"task-{i}-0"]}
+                    for i in range(3)],
+    }
+    return common, exp
+
+
+def _run_fedweit_rounds(ctx, common, exp):
+    from flreid_amd.runtime.builder import parser_clients, parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+    from flreid_amd.runtime.log import ExperimentLog
+    from flreid_amd.tools.utils import same_seeds
+
+    stage = ExperimentStage(common, [exp], ctx=ctx)
+    same_seeds(exp["random_seed"])
+    log = ExperimentLog(os.path.join(common["logs_dir"], "log.json"))
+    server = parser_server(exp, common)
+    client_names = [c["client_name"] for c in exp["clients"]]
+    owned = [i for i in range(len(client_names))
+             if ctx.owner_of(i) == ctx.rank]
+    clients = parser_clients(exp, common, owned_indices=owned)
+    by_name = {c.client_name: c for c in clients}
+    for r in (1, 2):
+        stage.process_one_round(r, server, by_name, client_names, exp, log)
+    state = server.model.model_state()
+    out = dict(state["sw"])
+    out.update({f"kb.{n}": p for n, p in state["aw_kb"].items()})
+    return {n: p.clone() for n, p in out.items()}
+
+
+def _worker_fedweit(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        common, exp = _fedweit_cfg(tmpdir, "dist")
+        state = _run_fedweit_rounds(ctx, common, exp)
+        with open(os.path.join(tmpdir, f"fedweit_state_r{rank}.pkl"), "wb") as f:
+            pickle.dump(state, f)
+    finally:
+        destroy_context()
+
+
+@pytest.mark.timeout(600)
+def test_fedweit_distributed_equals_single(tmp_path):
+    """FedWeIT 3 clients / 2 ranks: the C2 codec gather must reproduce the
+    single-process server (sw AND the sampled knowledge base — replica-
+    deterministic kb sampling), and BOTH rank replicas must agree."""
+    tmpdir = str(tmp_path)
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    from flreid_amd.parallel.comm import FedContext
+    common, exp = _fedweit_cfg(tmpdir, "single")
+    single = _run_fedweit_rounds(FedContext(), common, exp)
+
+    port = _free_port()
+    mp.spawn(_worker_fedweit, args=(2, port, tmpdir), nprocs=2, join=True)
+    states = []
+    for r in (0, 1):
+        with open(os.path.join(tmpdir, f"fedweit_state_r{r}.pkl"), "rb") as f:
+            states.append(pickle.load(f))
+    # replica agreement across ranks (the kb-sampling divergence trap)
+    assert set(states[0]) == set(states[1])
+    for n in states[0]:
+        assert torch.allclose(states[0][n], states[1][n], atol=0), n
+    # distributed == single-process
+    assert set(single) == set(states[0])
+    for n in single:
+        assert torch.allclose(single[n], states[0][n], atol=1e-6), n
