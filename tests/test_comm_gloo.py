@@ -400,3 +400,79 @@ def test_fedweit_distributed_equals_single(tmp_path):
     assert set(single) == set(states[0])
     for n in single:
         assert torch.allclose(single[n], states[0][n], atol=1e-6), n
+
+
+def _fedcurv_cfg(tmpdir, tag):
+    common, exp = _fedweit_cfg(tmpdir, tag)
+    exp = dict(exp)
+    exp["exp_name"] = "dist-fedcurv"
+    exp["exp_method"] = "fedcurv"
+    exp["model_opts"] = {"name": "resnet18", "num_classes": 64,
+                         "last_stride": 1, "neck": "bnneck",
+                         "lambda_penalty": 1.0,
+                         "fine_tuning": ["classifier"]}
+    return common, exp
+
+
+def _run_fedcurv_rounds(ctx, common, exp):
+    from flreid_amd.runtime.builder import parser_clients, parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+    from flreid_amd.runtime.log import ExperimentLog
+    from flreid_amd.tools.utils import same_seeds
+
+    stage = ExperimentStage(common, [exp], ctx=ctx)
+    same_seeds(exp["random_seed"])
+    log = ExperimentLog(os.path.join(common["logs_dir"], "log.json"))
+    server = parser_server(exp, common)
+    client_names = [c["client_name"] for c in exp["clients"]]
+    owned = [i for i in range(len(client_names))
+             if ctx.owner_of(i) == ctx.rank]
+    clients = parser_clients(exp, common, owned_indices=owned)
+    by_name = {c.client_name: c for c in clients}
+    for r in (1, 2):
+        stage.process_one_round(r, server, by_name, client_names, exp, log)
+    # the aggregated model + the full mesh every client would receive
+    out = {f"model.{n}": p.detach().clone()
+           for n, p in server.model.named_parameters() if p.requires_grad}
+    mesh = server.get_dispatch_incremental_state("client-0")
+    for i, (ps, fs) in enumerate(zip(mesh["other_clients_incremental_params"],
+                                     mesh["other_clients_precision_matrices"])):
+        for n, p in ps.items():
+            out[f"mesh{i}.p.{n}"] = p.clone()
+        for n, p in fs.items():
+            out[f"mesh{i}.f.{n}"] = p.clone()
+    return out
+
+
+def _worker_fedcurv(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        common, exp = _fedcurv_cfg(tmpdir, "dist")
+        state = _run_fedcurv_rounds(ctx, common, exp)
+        if rank == 0:
+            with open(os.path.join(tmpdir, "fedcurv_state.pkl"), "wb") as f:
+                pickle.dump(state, f)
+    finally:
+        destroy_context()
+
+
+@pytest.mark.timeout(600)
+def test_fedcurv_distributed_equals_single(tmp_path):
+    """FedCurv 3 clients / 2 ranks: params+Fisher full mesh (C3) through the
+    codec gather must match the single-process run exactly."""
+    tmpdir = str(tmp_path)
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    from flreid_amd.parallel.comm import FedContext
+    common, exp = _fedcurv_cfg(tmpdir, "single")
+    single = _run_fedcurv_rounds(FedContext(), common, exp)
+
+    port = _free_port()
+    mp.spawn(_worker_fedcurv, args=(2, port, tmpdir), nprocs=2, join=True)
+    with open(os.path.join(tmpdir, "fedcurv_state.pkl"), "rb") as f:
+        dist_state = pickle.load(f)
+    assert set(single) == set(dist_state)
+    for n in single:
+        assert torch.allclose(single[n], dist_state[n], atol=1e-6), n
