@@ -51,6 +51,13 @@ def parse_args():
     p.add_argument("--imgs-per-id", type=int, default=8,
                    help="train images per identity per task")
     p.add_argument("--num-classes", type=int, default=8000)
+    p.add_argument("--lambda-k", type=int, default=2000,
+                   help="FedSTIL exemplar budget.  The reference's ResNet-50 "
+                        "config uses 12000 over the thousands of identities "
+                        "of 5 real datasets (a few exemplars per id); scaled "
+                        "to this synthetic 64-id task it gives m=32 per id — "
+                        "a proportionally LARGER per-id rehearsal load")
+    p.add_argument("--lambda-l1", type=float, default=1e-4)
     p.add_argument("--cpu", action="store_true", help="debug on CPU")
     p.add_argument("--channels-last", dest="channels_last",
                    action="store_true", default=True,
@@ -81,8 +88,8 @@ def build_configs(args, world_size, rank):
                      "persist_comm_ckpts": False},
         "model_opts": {"name": args.model, "num_classes": args.num_classes,
                        "last_stride": 1, "neck": "bnneck",
-                       "atten_default": 0.9, "lambda_l1": 1e-4,
-                       "lambda_k": 2000,
+                       "atten_default": 0.9, "lambda_l1": args.lambda_l1,
+                       "lambda_k": args.lambda_k,
                        "fine_tuning": (["base.layers.3", "classifier"]
                                        if args.model.startswith("swin")
                                        else ["base.layer4", "classifier"])},
@@ -210,6 +217,9 @@ def main():
                 "img": args.img,
                 "images_per_round_per_client": imgs_per_client,
                 "num_classes": args.num_classes,
+                "lambda_k": args.lambda_k,
+                "rehearsal_set": (f"{args.lambda_k} exemplar budget + task "
+                                  f"(m={-(-args.lambda_k // args.ids)}/id)"),
                 "parallelism": f"fed-dp{max(1, ctx.world_size)} (1 client/GPU)",
             },
         }
