@@ -33,15 +33,21 @@ def needs_rebuild() -> bool:
     return False
 
 
-def build(force: bool = False, arch: str = "gfx950", verbose: bool = True) -> str:
+def build(force: bool = False, arch: str = "gfx950", verbose: bool = True,
+          debug: bool = False) -> str:
+    """debug=True (or --debug / FLREID_BUILD_DEBUG=1): -g -O1 plus host
+    address-sanitizer — the kernel-debug build for rocgdb / sanitizer runs
+    (SURVEY.md §5.2); release build is the default."""
     if not force and not needs_rebuild():
         if verbose:
             print(f"[flreid build] up to date: {OUT}")
         return OUT
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
     includes = [sysconfig.get_paths()["include"]] + _pybind11_includes()
+    debug = debug or os.environ.get("FLREID_BUILD_DEBUG", "0") == "1"
+    opt = ["-g", "-O1"] if debug else ["-O3"]
     cmd = [
-        hipcc, f"--offload-arch={arch}", "-O3", "-std=c++17", "-fPIC",
+        hipcc, f"--offload-arch={arch}", *opt, "-std=c++17", "-fPIC",
         "-shared", "-fvisibility=hidden",
         *[f"-I{p}" for p in includes],
         *[os.path.join(CSRC, s) for s in SOURCES],
@@ -54,4 +60,5 @@ def build(force: bool = False, arch: str = "gfx950", verbose: bool = True) -> st
 
 
 if __name__ == "__main__":
-    build(force="--force" in sys.argv)
+    build(force="--force" in sys.argv or "--debug" in sys.argv,
+          debug="--debug" in sys.argv)

@@ -228,9 +228,12 @@ class ExperimentStage:
     # -------------------------------------------------------------- workers
     @clear_cache
     def _process_train(self, client, log: ExperimentLog, curr_round: int) -> None:
+        import time as _time
+
         task = client.task_pipeline.next_task()
         if task["tr_epochs"] == 0:
             return
+        t0 = _time.perf_counter()
         tr_output = client.train(
             epochs=task["tr_epochs"],
             task_name=task["task_name"],
@@ -238,8 +241,16 @@ class ExperimentStage:
             val_loader=task["query_loader"],
             device=self.device,
         )
+        dt = _time.perf_counter() - t0
+        # per-round observability the reference lacked:
+        # wall clock + throughput of the north-star metric per client-round
+        rec = {"tr_acc": tr_output["accuracy"], "tr_loss": tr_output["loss"],
+               "tr_ms": round(dt * 1000.0, 1)}
+        n_imgs = int(tr_output.get("data_count", 0) or 0)
+        if n_imgs and dt > 0:
+            rec["tr_images_per_sec"] = round(n_imgs / dt, 1)
         log.record(f"data.{client.client_name}.{curr_round}.{task['task_name']}",
-                   {"tr_acc": tr_output["accuracy"], "tr_loss": tr_output["loss"]})
+                   rec)
 
     @clear_cache
     def _process_val(self, client, log: ExperimentLog, curr_round: int) -> None:
