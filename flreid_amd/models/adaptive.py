@@ -131,14 +131,31 @@ class AdaptiveConv2d(AdaptiveBase):
         if (data.is_cuda and gw.dim() == 4 and gw.shape[2] == 1
                 and gw.shape[3] == 1 and _is_one(self.stride)
                 and _is_zero(self.padding)):
-            # pointwise conv == GEMM over N·H·W: runs fwd AND bwd through
-            # hipBLASLt (MIOpen picked a naive weight-grad kernel for these
-            # ReID shapes — measured 2.4 ms/call); on channels-last input the
-            # [NHW, C] view is free
+            # pointwise conv == GEMM over N·H·W; on channels-last input the
+            # [NHW, C] view is free.  Two routes (bwd always hipBLASLt —
+            # MIOpen picked a naive weight-grad kernel for these shapes):
+            #  - OPT-IN (FLREID_FUSED_1X1=1) fused K2 MFMA GEMM with
+            #    compose-in-prologue (θ never hits HBM): the standalone probe
+            #    (benchmarks/adaptive_1x1_probe.py) shows wins at K≤512 /
+            #    K≤1024·N≤1024, but the end-to-end effect (~+1%) sits below
+            #    the box-to-box noise floor, so the proven path stays default;
+            #  - compose kernel + hipBLASLt otherwise.
             b, c, h, w = data.shape
+            n_out = gw.shape[0]
             xv = data.permute(0, 2, 3, 1).reshape(-1, c)
-            theta = self.composed_weight().view(gw.shape[0], c)
-            y = F.linear(xv, theta, self.adaptive_bias)
+            import os as _os
+            if (not self.global_weight_atten.requires_grad
+                    and not gw.requires_grad and c % 32 == 0
+                    and (c <= 512 or (c <= 1024 and n_out <= 1024))
+                    and _os.environ.get("FLREID_FUSED_1X1", "0") == "1"
+                    and ops.extension_available()):
+                y = ops.adaptive_linear(xv, gw.view(n_out, c),
+                                        self.global_weight_atten,
+                                        self.adaptive_weight.view(n_out, c),
+                                        self.adaptive_bias)
+            else:
+                theta = self.composed_weight().view(n_out, c)
+                y = F.linear(xv, theta, self.adaptive_bias)
             return y.view(b, h, w, -1).permute(0, 3, 1, 2)
         return F.conv2d(data, self.composed_weight(), self.adaptive_bias,
                         stride=self.stride, padding=self.padding)
