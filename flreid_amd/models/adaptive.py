@@ -149,8 +149,10 @@ class AdaptiveConv2d(AdaptiveBase):
                     and (c <= 512 or (c <= 1024 and n_out <= 1024))
                     and _os.environ.get("FLREID_FUSED_1X1", "0") == "1"
                     and ops.extension_available()):
-                y = ops.adaptive_linear(xv, gw.view(n_out, c),
-                                        self.global_weight_atten,
+                # the K2 kernel's contract is atten[K]: broadcast the conv's
+                # scalar atten (last dim of the 4-D weight is 1) to length K
+                atten_k = self.global_weight_atten.expand(c).contiguous()
+                y = ops.adaptive_linear(xv, gw.view(n_out, c), atten_k,
                                         self.adaptive_weight.view(n_out, c),
                                         self.adaptive_bias)
             else:

@@ -408,6 +408,7 @@ def adaptive_linear(x: torch.Tensor, gw: torch.Tensor, atten: torch.Tensor,
     gw/atten), eager compose+linear elsewhere."""
     if (x.is_cuda and x.dim() == 2 and not gw.requires_grad
             and not atten.requires_grad and gw.shape[-1] % 32 == 0
+            and atten.numel() == gw.shape[-1]      # kernel contract: atten[K]
             and extension_available()):
         return _AdaptiveLinearFn.apply(x, gw, atten, aw, bias)
     theta = adaptive_compose(gw, atten, aw)

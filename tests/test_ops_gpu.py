@@ -447,3 +447,34 @@ def test_l1_drift_fused_matches_foreach():
     # cached-metadata path: second call must reuse tables and stay correct
     loss2 = ops.l1_drift(pairs)
     assert torch.allclose(loss2, ref_loss, rtol=1e-6)
+
+
+def test_adaptive_conv1x1_fused_route(monkeypatch):
+    """Opt-in FLREID_FUSED_1X1=1 pointwise route (K2 GEMM with scalar atten,
+    compose-in-prologue) vs the compose+hipBLASLt default."""
+    import copy
+    from flreid_amd.models.adaptive import AdaptiveConv2d
+    torch.manual_seed(4)
+    w = torch.randn(512, 1024, 1, 1, device="cuda") * 0.02
+    conv = AdaptiveConv2d(global_weight=w, stride=1, padding=0,
+                          atten_default=0.9).cuda()
+    conv2 = copy.deepcopy(conv)
+    x = (torch.randn(4, 1024, 8, 4, device="cuda").bfloat16()
+         .to(memory_format=torch.channels_last).requires_grad_(True))
+    x2 = x.detach().clone().requires_grad_(True)
+
+    # both routes run under bf16 autocast, like the training round
+    monkeypatch.setenv("FLREID_FUSED_1X1", "1")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = conv(x)
+    monkeypatch.setenv("FLREID_FUSED_1X1", "0")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y2 = conv2(x2)
+    assert torch.allclose(y.float(), y2.float(), atol=0.1, rtol=5e-2)
+
+    dy = torch.randn_like(y2.float())
+    y.backward(dy.to(y.dtype))
+    y2.backward(dy.to(y2.dtype))
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=0.1, rtol=5e-2)
+    assert torch.allclose(conv.adaptive_weight.grad,
+                          conv2.adaptive_weight.grad, atol=0.05, rtol=5e-2)
