@@ -476,3 +476,70 @@ def test_fedcurv_distributed_equals_single(tmp_path):
     assert set(single) == set(dist_state)
     for n in single:
         assert torch.allclose(single[n], dist_state[n], atol=1e-6), n
+
+
+def _atten_cfg(tmpdir, tag):
+    common, exp = _fedstil_cfg(tmpdir, tag)
+    exp = dict(exp)
+    exp["exp_name"] = "dist-fedstil-atten"
+    exp["exp_method"] = "fedstil-atten"
+    exp["model_opts"] = dict(exp["model_opts"])
+    exp["model_opts"]["atten_default"] = 0.0
+    return common, exp
+
+
+def _run_atten_rounds(ctx, common, exp):
+    from flreid_amd.runtime.builder import parser_clients, parser_server
+    from flreid_amd.runtime.experiment import ExperimentStage
+    from flreid_amd.runtime.log import ExperimentLog
+    from flreid_amd.tools.utils import same_seeds
+
+    stage = ExperimentStage(common, [exp], ctx=ctx)
+    same_seeds(exp["random_seed"])
+    log = ExperimentLog(os.path.join(common["logs_dir"], "log.json"))
+    server = parser_server(exp, common)
+    client_names = [c["client_name"] for c in exp["clients"]]
+    owned = [i for i in range(len(client_names))
+             if ctx.owner_of(i) == ctx.rank]
+    clients = parser_clients(exp, common, owned_indices=owned)
+    by_name = {c.client_name: c for c in clients}
+    for r in (1, 2, 3):
+        stage.process_one_round(r, server, by_name, client_names, exp, log)
+    state = server.model.model_state()["global_weight"]
+    return {n: p.clone() for n, p in state.items()}
+
+
+def _worker_atten(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        common, exp = _atten_cfg(tmpdir, "dist")
+        state = _run_atten_rounds(ctx, common, exp)
+        if rank == 0:
+            with open(os.path.join(tmpdir, "atten_state.pkl"), "wb") as f:
+                pickle.dump(state, f)
+    finally:
+        destroy_context()
+
+
+@pytest.mark.timeout(600)
+def test_fedstil_atten_distributed_equals_single(tmp_path):
+    """fedstil-atten: the server CONCATENATES uploads into a growing stack —
+    the distributed gather must reproduce the single-process stack exactly
+    (same slot order, same values) under partial participation."""
+    tmpdir = str(tmp_path)
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    from flreid_amd.parallel.comm import FedContext
+    common, exp = _atten_cfg(tmpdir, "single")
+    single = _run_atten_rounds(FedContext(), common, exp)
+
+    port = _free_port()
+    mp.spawn(_worker_atten, args=(2, port, tmpdir), nprocs=2, join=True)
+    with open(os.path.join(tmpdir, "atten_state.pkl"), "rb") as f:
+        dist_state = pickle.load(f)
+    assert set(single) == set(dist_state)
+    for n in single:
+        assert single[n].shape == dist_state[n].shape, n
+        assert torch.allclose(single[n], dist_state[n], atol=1e-6), n
