@@ -20,6 +20,11 @@ def test_fedavg_e2e(tiny_common, tiny_exp_config, tmp_path, monkeypatch):
     r2 = data["client-0"]["2"]
     any_task = next(iter(r2.values()))
     assert "val_map" in any_task or "tr_acc" in any_task
+    # per-round perf observability
+    train_recs = [v for rnd in data["client-0"].values() for v in rnd.values()
+                  if "tr_acc" in v]
+    assert train_recs and all("tr_ms" in v for v in train_recs)
+    assert any("tr_images_per_sec" in v for v in train_recs)
 
     # ckpt audit layout (ref:experiment.py:199-202,233-241 + ckpts/README.md)
     server_dir = os.path.join(tiny_common["checkpoints_dir"], "tiny", "server")

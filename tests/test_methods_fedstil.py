@@ -114,3 +114,63 @@ def test_herding_budget():
         "atten_default": 0.9, "lambda_k": 10})
     model.ids.update([1, 2, 3])
     assert model.m == math.ceil(10 / 3)
+
+
+def test_exemplar_store_across_task_switch(tmp_path, monkeypatch):
+    """Across a task switch the id set grows, m shrinks, and the store must
+    carry exemplars for BOTH tasks' identities with ≤ m entries each
+    (ref:methods/fedstil.py:349-399 + reduce at train start)."""
+    monkeypatch.chdir(tmp_path)
+    from flreid_amd.runtime.builder import parser_clients
+
+    common = {
+        "datasets_dir": "synthetic://ids=4,train=3,query=2,gallery=2,hw=32x16,idspace=64",
+        "checkpoints_dir": str(tmp_path / "ck"),
+        "logs_dir": str(tmp_path / "lg"),
+        "parallel": 1, "device": ["cpu"], "defaults": {},
+    }
+    exp = {
+        "exp_name": "switch", "exp_method": "fedstil", "random_seed": 3,
+        "exp_opts": {"comm_rounds": 2, "val_interval": 0, "online_clients": 1},
+        "model_opts": {"name": "resnet18", "num_classes": 64, "last_stride": 1,
+                       "neck": "bnneck", "atten_default": 0.9,
+                       "lambda_l1": 1e-4, "lambda_k": 8,
+                       "fine_tuning": ["classifier"]},
+        "criterion_opts": {"name": "cross_entropy", "num_classes": 64,
+                           "epsilon": 0.1},
+        "optimizer_opts": {"name": "adam", "lr": 1e-3, "weight_decay": 1e-5},
+        "scheduler_opts": {"name": "step_lr", "step_size": 5},
+        "task_opts": {"sustain_rounds": 1, "train_epochs": 1,
+                      "augment_opts": {"level": "none", "img_size": [32, 16],
+                                       "norm_mean": [0.485, 0.456, 0.406],
+                                       "norm_std": [0.229, 0.224, 0.225]},
+                      "loader_opts": {"batch_size": 4, "num_workers": 0,
+                                      "pin_memory": False,
+                                      "persistent_workers": False,
+                                      "multiprocessing_context": None}},
+        "server": {"server_name": "server", "distance_calculate_step": 10,
+                   "distance_calculate_decay": 0.8},
+        "clients": [{"client_name": "client-0",
+                     "tasks": ["task-0-0", "task-0-1"]}],
+    }
+    client = parser_clients(exp, common)[0]
+
+    # task 1
+    t1 = client.task_pipeline.next_task()
+    client.train(epochs=1, task_name=t1["task_name"],
+                 tr_loader=t1["tr_loader"], val_loader=t1["query_loader"],
+                 device="cpu")
+    ids_t1 = set(client.model.examplars.keys())
+    assert ids_t1 and all(
+        p.shape[0] <= client.model.m for p, _c in client.model.examplars.values())
+
+    # task 2 (different identities via the synthetic per-task id stride)
+    t2 = client.task_pipeline.next_task()
+    assert t2["task_name"] != t1["task_name"]
+    client.train(epochs=1, task_name=t2["task_name"],
+                 tr_loader=t2["tr_loader"], val_loader=t2["query_loader"],
+                 device="cpu")
+    ids_all = set(client.model.examplars.keys())
+    m = client.model.m
+    assert ids_t1 <= ids_all and len(ids_all) > len(ids_t1)
+    assert all(p.shape[0] <= m for p, _c in client.model.examplars.values())
