@@ -34,6 +34,33 @@ os.environ.setdefault("FLREID_FAST_CONV", "1")
 os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 os.environ.setdefault("FLREID_GPU_AUGMENT", "1")
 
+
+def _seed_miopen_db():
+    """Seed MIOPEN_USER_DB_PATH from the shipped pre-tuned find DB
+    (flreid_amd/data/miopen_udb/) so every box starts from the same conv-algo
+    picks — FIND_MODE=FAST occasionally landed on im2col/naive fallbacks for
+    the eval-path shapes, a 10 %+ box lottery.  MIOpen writes to the user DB
+    path, so the files are copied somewhere writable first."""
+    if "MIOPEN_USER_DB_PATH" in os.environ:
+        return
+    src = os.path.join(REPO_ROOT, "flreid_amd", "data", "miopen_udb")
+    if not os.path.isdir(src):
+        return
+    import shutil
+    import tempfile
+    dst = os.path.join(tempfile.gettempdir(), "flreid_miopen_udb")
+    os.makedirs(dst, exist_ok=True)
+    for f in os.listdir(src):
+        if not f.endswith(".txt"):
+            continue
+        t = os.path.join(dst, f)
+        if not os.path.exists(t):
+            shutil.copy(os.path.join(src, f), t)
+    os.environ["MIOPEN_USER_DB_PATH"] = dst
+
+
+_seed_miopen_db()
+
 import torch  # noqa: E402
 
 
