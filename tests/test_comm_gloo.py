@@ -543,3 +543,38 @@ def test_fedstil_atten_distributed_equals_single(tmp_path):
     for n in single:
         assert single[n].shape == dist_state[n].shape, n
         assert torch.allclose(single[n], dist_state[n], atol=1e-6), n
+
+
+def _worker_codec_bf16(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    os.environ["FLREID_COMM_DTYPE"] = "bf16"
+    try:
+        from flreid_amd.parallel.codec import sync_client_states
+        from flreid_amd.parallel.comm import destroy_context, init_context
+        ctx = init_context(device="cpu")
+        try:
+            state = {
+                "train_cnt": 10 + rank,
+                "task_token": torch.full((4,), float(rank)),
+                "incremental_sw": {"w": torch.full((3, 2), float(rank + 1))},
+            }
+            merged = sync_client_states(ctx, {f"client-{rank}": state})
+            assert set(merged) == {"client-0", "client-1"}
+            for r in (0, 1):
+                s = merged[f"client-{r}"]
+                # bf16 wire exactly represents these small values
+                assert torch.allclose(s["incremental_sw"]["w"],
+                                      torch.full((3, 2), float(r + 1)))
+                assert s["incremental_sw"]["w"].dtype == torch.float32
+        finally:
+            destroy_context()
+    finally:
+        os.environ.pop("FLREID_COMM_DTYPE", None)
+
+
+def test_tensor_codec_gather_bf16_wire(tmp_path):
+    """FLREID_COMM_DTYPE=bf16 halves wire bytes; states come back in their
+    original dtype."""
+    port = _free_port()
+    mp.spawn(_worker_codec_bf16, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
