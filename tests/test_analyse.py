@@ -45,3 +45,27 @@ def test_grad_cam_runs():
     cam = grad_cam(net, net.base.layer4[-1], img)
     assert cam.shape == (64, 32)
     assert float(cam.min()) >= 0.0 and float(cam.max()) <= 1.0
+
+
+def test_analyse_consumes_real_run_log(tiny_common, tiny_exp_config, tmp_path,
+                                       monkeypatch):
+    """Schema integration: the analysis layer must read the JSON an actual
+    experiment run writes (guards against log-schema drift)."""
+    import json
+    import os
+
+    from flreid_amd.analyse.accuracy import load_log
+    from flreid_amd.parallel.comm import FedContext
+    from flreid_amd.runtime.experiment import ExperimentStage
+
+    monkeypatch.chdir(tmp_path)
+    stage = ExperimentStage(tiny_common, [tiny_exp_config], ctx=FedContext())
+    stage.run_experiment(tiny_exp_config)
+    logs = os.listdir(tiny_common["logs_dir"])
+    assert len(logs) == 1
+    records = load_log(os.path.join(tiny_common["logs_dir"], logs[0]))
+
+    curve = accuracy_on_round(records, "val_rank_1")
+    assert curve and all(0.0 <= v <= 1.0 for v in curve.values())
+    forg = mean_forgetting(records, "val_rank_1")
+    assert isinstance(forg, float)
