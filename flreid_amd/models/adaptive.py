@@ -22,6 +22,7 @@ dims but the last.
 
 from __future__ import annotations
 
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -143,11 +144,10 @@ class AdaptiveConv2d(AdaptiveBase):
             b, c, h, w = data.shape
             n_out = gw.shape[0]
             xv = data.permute(0, 2, 3, 1).reshape(-1, c)
-            import os as _os
             if (not self.global_weight_atten.requires_grad
                     and not gw.requires_grad and c % 32 == 0
                     and (c <= 512 or (c <= 1024 and n_out <= 1024))
-                    and _os.environ.get("FLREID_FUSED_1X1", "0") == "1"
+                    and os.environ.get("FLREID_FUSED_1X1", "0") == "1"
                     and ops.extension_available()):
                 # the K2 kernel's contract is atten[K]: broadcast the conv's
                 # scalar atten (last dim of the 4-D weight is 1) to length K

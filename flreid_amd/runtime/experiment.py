@@ -140,15 +140,16 @@ class ExperimentStage:
         # server deterministic), but the potentially expensive dispatch state
         # (e.g. FedSTIL's personalized mixture) is only computed where it is
         # consumed: on the owning rank, or on rank 0 for the ckpt audit trail
+        from flreid_amd.runtime.hipgraph import (dump_phases, phase,
+                                                 phase_timers_enabled)
         for cname in online:
             first_contact = cname not in server.clients
             if first_contact:
                 server.register_client(cname)
-            from flreid_amd.runtime.hipgraph import phase as _ph
             needed = (cname in by_name) or (persist_comm and self.ctx.is_rank0())
             dispatch_state = None
             if needed:
-              with _ph("dispatch"):
+              with phase("dispatch"):
                 if first_contact:
                     dispatch_state = server.get_dispatch_integrated_state(cname)
                     if dispatch_state is not None and cname in by_name:
@@ -173,8 +174,7 @@ class ExperimentStage:
             if cname in by_name:
                 same_seeds((base_seed * 1000003 + curr_round * 1009 +
                             client_names.index(cname)) % (2 ** 31))
-                from flreid_amd.runtime.hipgraph import phase as _ph3
-                with _ph3("train_total"):
+                with phase("train_total"):
                     self._process_train(by_name[cname], log, curr_round)
 
         # ---- validation every val_interval rounds --------------------------
@@ -188,8 +188,7 @@ class ExperimentStage:
             if cname not in by_name:
                 continue
             client = by_name[cname]
-            from flreid_amd.runtime.hipgraph import phase as _ph2
-            with _ph2("upload_build"):
+            with phase("upload_build"):
                 state = client.get_incremental_state()
             if persist_comm:
                 client.save_state(f"{curr_round}-{cname}-{server.server_name}",
@@ -202,7 +201,6 @@ class ExperimentStage:
 
         # ---- sync uploads across ranks (ONE gather per round) --------------
         from flreid_amd.parallel.codec import sync_client_states
-        from flreid_amd.runtime.hipgraph import dump_phases, phase, phase_timers_enabled
         collective = getattr(server, "collective_aggregate", None)
         if collective is not None and self.ctx.is_distributed:
             # C1 fast path: pre-scaled all-reduce, no upload replication
