@@ -44,7 +44,7 @@ def _materialize(node, rng):
     return node
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=60, deadline=None, derandomize=True)
 @given(skeleton=states(), seed=st.integers(0, 2 ** 16))
 def test_flatten_unflatten_roundtrip(skeleton, seed):
     import numpy as np
