@@ -69,3 +69,26 @@ def test_analyse_consumes_real_run_log(tiny_common, tiny_exp_config, tmp_path,
     assert curve and all(0.0 <= v <= 1.0 for v in curve.values())
     forg = mean_forgetting(records, "val_rank_1")
     assert isinstance(forg, float)
+
+
+def test_plots_render(tmp_path):
+    """Round-curve and forgetting plots render to PNG (matplotlib is in the
+    image; the functions still return None gracefully without it)."""
+    import json
+
+    from flreid_amd.analyse.accuracy import plot_accuracy_curves
+    from flreid_amd.analyse.forgetting import plot_forgetting
+
+    log_path = tmp_path / "run.json"
+    log_path.write_text(json.dumps(RECORDS))
+
+    p1 = plot_accuracy_curves([str(log_path)], "val_rank_1",
+                              str(tmp_path / "acc.png"))
+    p2 = plot_forgetting([str(log_path)], "val_rank_1",
+                         str(tmp_path / "forg.png"))
+    import importlib.util
+    if importlib.util.find_spec("matplotlib") is None:
+        assert p1 is None and p2 is None
+    else:
+        import os
+        assert os.path.getsize(p1) > 0 and os.path.getsize(p2) > 0
