@@ -71,3 +71,36 @@ def test_batched_augment_matches_distribution():
     out = aug_d.apply_batch(torch.rand(8, 3, 16, 8))
     assert out.shape == (8, 3, 16, 8)
     assert torch.isfinite(out).all()
+
+
+def test_augmentation_levels_shapes_and_semantics():
+    """All five reference levels (ref:datasets/image_augmentation.py:6-71):
+    shared output shape, 'none' deterministic, erasing levels perturb."""
+    import torch
+    from flreid_amd.data.augment import (augmentation_default,
+                                         augmentation_drastic,
+                                         augmentation_none,
+                                         augmentation_rose,
+                                         augmentation_sharp)
+
+    torch.manual_seed(0)
+    img = torch.rand(3, 40, 20)
+    size = (32, 16)
+    levels = {
+        "none": augmentation_none(size), "default": augmentation_default(size),
+        "rose": augmentation_rose(size), "sharp": augmentation_sharp(size),
+        "drastic": augmentation_drastic(size),
+    }
+    for name, aug in levels.items():
+        out = aug(img.clone())
+        assert out.shape == (3, *size), name
+
+    a = augmentation_none(size)(img.clone())
+    b = augmentation_none(size)(img.clone())
+    assert torch.equal(a, b)                   # deterministic
+
+    # drastic erases with p=0.9: over 50 draws some output must differ
+    torch.manual_seed(1)
+    dr = augmentation_drastic(size)
+    outs = [dr(img.clone()) for _ in range(50)]
+    assert any(not torch.equal(outs[0], o) for o in outs[1:])
