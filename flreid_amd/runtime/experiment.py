@@ -110,6 +110,8 @@ class ExperimentStage:
         log.sync(self.ctx)
 
         comm_rounds = int(exp_config["exp_opts"]["comm_rounds"])
+        from flreid_amd.runtime.hipgraph import suggest_epoch_graph
+        suggest_epoch_graph(comm_rounds)
         for curr_round in range(1, comm_rounds + 1):
             if self.ctx.is_rank0():
                 self.logger.info(f"Start communication round: {curr_round:03d}/{comm_rounds:03d}")

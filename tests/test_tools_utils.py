@@ -56,3 +56,24 @@ def test_trainable_params_filters():
         p.requires_grad_(False)
     names = set(trainable_params(m).keys())
     assert names == {"1.weight", "1.bias"}
+
+
+def test_epoch_graph_auto_threshold(monkeypatch):
+    """Long runs auto-enable whole-epoch capture; explicit env always wins
+    (runtime/hipgraph.py)."""
+    from flreid_amd.runtime import hipgraph as hg
+
+    monkeypatch.delenv("FLREID_EPOCH_GRAPH", raising=False)
+    monkeypatch.setattr(hg.torch.cuda, "is_available", lambda: True)
+
+    hg.suggest_epoch_graph(60)
+    assert not hg.epoch_graph_enabled()
+    hg.suggest_epoch_graph(500)
+    assert hg.epoch_graph_enabled()
+
+    monkeypatch.setenv("FLREID_EPOCH_GRAPH", "0")
+    assert not hg.epoch_graph_enabled()          # explicit off wins
+    monkeypatch.setenv("FLREID_EPOCH_GRAPH", "1")
+    hg.suggest_epoch_graph(5)
+    assert hg.epoch_graph_enabled()              # explicit on wins
+    hg.suggest_epoch_graph(0)                    # reset module state
