@@ -77,3 +77,34 @@ def test_epoch_graph_auto_threshold(monkeypatch):
     hg.suggest_epoch_graph(5)
     assert hg.epoch_graph_enabled()              # explicit on wins
     hg.suggest_epoch_graph(0)                    # reset module state
+
+
+def test_fedcontext_single_process_collectives():
+    """World-size-1 FedContext: every collective is the identity
+    (parallel/comm.py — the reference-simulator-compatible mode)."""
+    from flreid_amd.parallel.comm import FedContext
+
+    ctx = FedContext()
+    assert not ctx.is_distributed and ctx.is_rank0()
+    assert [ctx.owner_of(i) for i in range(4)] == [0, 0, 0, 0]
+    assert ctx.all_gather_object({"a": 1}) == [{"a": 1}]
+    assert ctx.broadcast_object(5) == 5
+    assert ctx.all_reduce_scalar(2.5) == 2.5
+    flat = ctx.all_gather_flat(torch.arange(3.0))
+    assert flat.shape == (1, 3) and torch.equal(flat[0], torch.arange(3.0))
+    out = ctx.weighted_allreduce({"w": torch.ones(2)}, 0.5)
+    assert torch.allclose(out["w"], torch.full((2,), 0.5))
+
+
+def test_synthetic_uri_parsing():
+    from flreid_amd.data.synthetic import parse_synthetic_dir
+
+    opts = parse_synthetic_dir(
+        "synthetic://ids=8,train=4,query=2,gallery=3,hw=64x32,idspace=128")
+    assert opts["ids"] == 8 and opts["train"] == 4
+    assert opts["shape"] == (3, 64, 32) and opts["idspace"] == 128
+    # defaults fill unspecified keys
+    d = parse_synthetic_dir("synthetic://ids=2")
+    assert d["ids"] == 2 and "train" in d and d["shape"] == (3, 128, 64)
+    # non-synthetic paths decline
+    assert parse_synthetic_dir("/data/reid") is None
