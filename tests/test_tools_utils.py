@@ -108,3 +108,36 @@ def test_synthetic_uri_parsing():
     assert d["ids"] == 2 and "train" in d and d["shape"] == (3, 128, 64)
     # non-synthetic paths decline
     assert parse_synthetic_dir("/data/reid") is None
+
+
+def test_async_ckpt_writer_ordering(tmp_path):
+    """runtime/io.py: same-path writes apply in submission order; flush
+    drains; the snapshot decouples from later mutation."""
+    from flreid_amd.runtime.io import AsyncCkptWriter
+
+    w = AsyncCkptWriter()
+    path = str(tmp_path / "s.ckpt")
+    live = {"x": torch.zeros(4)}
+    w.submit(path, live)
+    live["x"].fill_(1.0)          # mutation AFTER submit must not leak
+    w.flush()
+    assert torch.equal(torch.load(path, weights_only=False)["x"],
+                       torch.zeros(4))
+
+    for i in range(5):
+        w.submit(path, {"i": torch.full((2,), float(i))})
+    w.flush()
+    assert torch.equal(torch.load(path, weights_only=False)["i"],
+                       torch.full((2,), 4.0))
+
+
+def test_save_ckpt_async_env(tmp_path, monkeypatch):
+    """save_ckpt routes through the async writer when FLREID_ASYNC_CKPT=1
+    and before_ckpt_read drains it."""
+    from flreid_amd.runtime.io import before_ckpt_read, save_ckpt
+
+    monkeypatch.setenv("FLREID_ASYNC_CKPT", "1")
+    p = str(tmp_path / "a.ckpt")
+    save_ckpt(p, {"v": torch.ones(3)})
+    before_ckpt_read()
+    assert torch.equal(torch.load(p, weights_only=False)["v"], torch.ones(3))
