@@ -1,0 +1,57 @@
+"""Criterion-class matrix tests (ref:criterions/ — CPU reference paths)."""
+
+import pytest
+import torch
+
+from flreid_amd.criterions import criterions
+from flreid_amd.criterions.triplet_loss import TripletLoss
+
+
+def _batch(n=12, d=16, ids=3, seed=0):
+    torch.manual_seed(seed)
+    feat = torch.randn(n, d, requires_grad=True)
+    target = torch.arange(n) % ids
+    return feat, target
+
+
+@pytest.mark.parametrize("norm_feat", [False, True])
+@pytest.mark.parametrize("hard_mining", [False, True])
+@pytest.mark.parametrize("margin", [0.3, None])
+def test_triplet_mode_matrix(norm_feat, hard_mining, margin):
+    """All mining×distance×margin combinations produce a finite scalar with
+    gradients (ref:criterions/triplet_loss.py:12-127)."""
+    feat, target = _batch()
+    crit = TripletLoss(margin=margin, norm_feat=norm_feat,
+                       hard_mining=hard_mining)
+    loss = crit(feature=feat, target=target, score=None)
+    assert loss.dim() == 0 and torch.isfinite(loss)
+    loss.backward()
+    assert feat.grad is not None and torch.isfinite(feat.grad).all()
+
+
+def test_registry_names():
+    """Registry exposes the reference's criteria (+ distill_kl, which the
+    reference shipped unregistered — SURVEY.md §2.5)."""
+    for name in ("cross_entropy", "triplet", "distill_kl"):
+        assert name in criterions, name
+
+
+def test_label_smooth_ce_class():
+    ce = criterions["cross_entropy"](num_classes=8, epsilon=0.1)
+    score = torch.randn(6, 8, requires_grad=True)
+    target = torch.randint(0, 8, (6,))
+    loss = ce(score=score, target=target, feature=None)
+    assert torch.isfinite(loss)
+    loss.backward()
+    # label smoothing: even a perfect prediction keeps nonzero loss
+    perfect = torch.full((2, 8), -20.0)
+    perfect[0, 3] = perfect[1, 5] = 20.0
+    l2 = ce(score=perfect, target=torch.tensor([3, 5]), feature=None)
+    assert l2 > 0
+
+
+def test_autocast_cpu_is_noop():
+    from flreid_amd.runtime.precision import autocast
+    with autocast("cpu"):
+        x = torch.randn(2, 2) @ torch.randn(2, 2)
+    assert x.dtype == torch.float32
