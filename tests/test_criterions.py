@@ -32,7 +32,7 @@ def test_triplet_mode_matrix(norm_feat, hard_mining, margin):
 def test_registry_names():
     """Registry exposes the reference's criteria (+ distill_kl, which the
     reference shipped unregistered — SURVEY.md §2.5)."""
-    for name in ("cross_entropy", "triplet", "distill_kl"):
+    for name in ("cross_entropy", "triplet_loss", "distill_kl"):
         assert name in criterions, name
 
 
