@@ -92,3 +92,21 @@ def test_plots_render(tmp_path):
     else:
         import os
         assert os.path.getsize(p1) > 0 and os.path.getsize(p2) > 0
+
+
+def test_per_task_grid_and_merged_plots(tmp_path):
+    import json
+    import os
+
+    from flreid_amd.analyse.accuracy import plot_merged, plot_per_task_grid
+
+    log_path = tmp_path / "run.json"
+    log_path.write_text(json.dumps(RECORDS))
+    p1 = plot_per_task_grid(str(log_path), "val_map",
+                            str(tmp_path / "grid.png"))
+    p2 = plot_merged([str(log_path)], str(tmp_path / "merged.png"))
+    import importlib.util
+    if importlib.util.find_spec("matplotlib") is None:
+        assert p1 is None and p2 is None
+    else:
+        assert os.path.getsize(p1) > 0 and os.path.getsize(p2) > 0
