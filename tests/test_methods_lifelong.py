@@ -77,3 +77,37 @@ def test_fedprox_e2e(tiny_common, tiny_exp_config, tmp_path, monkeypatch):
     log = stage.run_experiment(cfg)
     r1 = log.records["data"]["client-0"].get("1", {})
     assert any("tr_acc" in v for v in r1.values())
+
+
+def test_importance_values_match_manual():
+    """Golden math: Fisher = Σ_batches g²·(|batch|/n_batches) for MAS/EWC
+    accumulation (methods/_importance.py; ref:methods/ewc.py:56-78 scale)."""
+    from flreid_amd.methods import methods
+    torch.manual_seed(1)
+
+    class Op:
+        @staticmethod
+        def _invoke_train(model, data, target):
+            return {"loss": (model.net(data) ** 2).sum()}
+
+    net = nn.Linear(3, 2, bias=False)
+    w0 = net.weight.detach().clone()
+    data = torch.randn(4, 3)
+    loader = [(data, torch.zeros(4, dtype=torch.long), torch.zeros(4))]
+
+    mas = methods["mas"].Model(net=net, operator=Op())
+    mas.remember_task("t0", loader)
+
+    # manual: single batch -> scale = len(batch)/n_batches = 4/1
+    w = w0.clone().requires_grad_(True)
+    loss = ((data @ w.t()) ** 2).sum()
+    loss.backward()
+    expected = w.grad.abs() * 4.0           # MAS: |g|·scale
+    got = mas.precision_matrices["weight"]
+    assert torch.allclose(got, expected, atol=1e-5), (got, expected)
+
+    # penalty = λ·Σ F·(p−p_old)² — zero right after calculate, positive after drift
+    assert float(mas.penalty()) == 0.0
+    with torch.no_grad():
+        net.weight += 0.1
+    assert float(mas.penalty()) > 0.0
