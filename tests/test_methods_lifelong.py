@@ -111,3 +111,25 @@ def test_importance_values_match_manual():
     with torch.no_grad():
         net.weight += 0.1
     assert float(mas.penalty()) > 0.0
+
+
+def test_fedcurv_cross_client_penalty_manual():
+    """FedCurv penalty = λ·[own-EWC + Σ_other F_o·(p−p_o)²]
+    (ref:methods/fedcurv.py:79-86)."""
+    from flreid_amd.methods import methods
+    torch.manual_seed(2)
+
+    class Op:
+        @staticmethod
+        def _invoke_train(model, data, target):
+            return {"loss": (model.net(data) ** 2).sum()}
+
+    net = nn.Linear(3, 2, bias=False)
+    m = methods["fedcurv"].Model(net=net, operator=Op(), lambda_penalty=2.0)
+    # no remembered tasks -> own term zero; inject one other client
+    f_o = {"weight": torch.rand(2, 3)}
+    p_o = {"weight": net.weight.detach() + 0.5}
+    m.other_precision_matrices = [(f_o, p_o)]
+    pen = m.penalty()
+    expected = 2.0 * (f_o["weight"] * 0.25).sum()
+    assert torch.allclose(pen, expected, atol=1e-6)
