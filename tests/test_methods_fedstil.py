@@ -174,3 +174,59 @@ def test_exemplar_store_across_task_switch(tmp_path, monkeypatch):
     m = client.model.m
     assert ids_t1 <= ids_all and len(ids_all) > len(ids_t1)
     assert all(p.shape[0] <= m for p, _c in client.model.examplars.values())
+
+
+def test_personalized_dispatch_mixture_manual(tmp_path, monkeypatch):
+    """Golden math for the KL-token personalized dispatch: softmax over
+    normalized inverse decayed-KL distances, self weighted at the mean
+    (ref:methods/fedstil.py:1118-1164)."""
+    import math as _math
+
+    import torch.nn.functional as F
+    from flreid_amd.methods import methods
+    from flreid_amd.runtime.builder import parser_model
+    from flreid_amd.tools.distance import compute_kl_distance
+
+    monkeypatch.chdir(tmp_path)
+    fedstil = methods["fedstil"]
+    model = parser_model("fedstil", {
+        "name": "resnet18", "num_classes": 8, "last_stride": 1,
+        "neck": "bnneck", "fine_tuning": ["classifier"],
+        "atten_default": 0.9, "lambda_k": 8})
+    class _Op:
+        logger = None
+
+    server = fedstil.Server("server", model, _Op(), str(tmp_path),
+                            distance_calculate_step=1,
+                            distance_calculate_decay=0.8)
+
+    torch.manual_seed(5)
+    uploads = {}
+    for c in ("client-0", "client-1", "client-2"):
+        server.register_client(c)
+        state = {
+            "train_cnt": 4,
+            "task_token": torch.randn(16),
+            "incremental_sw": {"w": torch.randn(3)},
+        }
+        server.set_client_incremental_state(c, state)
+        uploads[c] = state
+
+    out = server.get_dispatch_incremental_state("client-0")["incremental_shared_params"]
+
+    # manual mixture
+    own = uploads["client-0"]["task_token"].unsqueeze(0)
+    names, inv = [], []
+    for c in ("client-1", "client-2"):
+        toks = server.token_memory[c][::-1]
+        dis = 1e-8
+        for cnt, t in enumerate(toks):
+            dis += float(compute_kl_distance(own, t.unsqueeze(0))) / (0.8 ** cnt)
+        names.append(c)
+        inv.append(1.0 / dis)
+    names.append("client-0")
+    inv.append(sum(inv) / len(inv))
+    w = torch.tensor([d / sum(inv) for d in inv]).softmax(dim=0)
+    expected = sum(uploads[c]["incremental_sw"]["w"] * float(m)
+                   for c, m in zip(names, w))
+    assert torch.allclose(out["w"], expected, atol=1e-6)
