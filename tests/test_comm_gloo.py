@@ -578,3 +578,32 @@ def test_tensor_codec_gather_bf16_wire(tmp_path):
     port = _free_port()
     mp.spawn(_worker_codec_bf16, args=(2, port, str(tmp_path)), nprocs=2,
              join=True)
+
+
+def _worker_codec_ragged(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.codec import sync_client_states
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        # rank-dependent tensor sizes -> unequal strides -> object fallback
+        state = {
+            "train_cnt": rank,
+            "incremental_sw": {"w": torch.full((2 + rank, 3), float(rank))},
+        }
+        merged = sync_client_states(ctx, {f"client-{rank}": state})
+        assert set(merged) == {"client-0", "client-1"}
+        for r in (0, 1):
+            w = merged[f"client-{r}"]["incremental_sw"]["w"]
+            assert w.shape == (2 + r, 3)
+            assert torch.allclose(w, torch.full((2 + r, 3), float(r)))
+    finally:
+        destroy_context()
+
+
+def test_tensor_codec_ragged_fallback(tmp_path):
+    """Uneven per-rank payloads must fall back to the object gather and
+    still merge every client's state (parallel/codec.py:114-119)."""
+    port = _free_port()
+    mp.spawn(_worker_codec_ragged, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
