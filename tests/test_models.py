@@ -51,3 +51,35 @@ def test_bnneck_classifier_has_no_bias():
     net = nets["resnet18"](num_classes=8, neck="bnneck")
     assert net.classifier.bias is None
     assert net.bottleneck.bias.requires_grad is False
+
+
+def test_staged_execution_matches_full_forward():
+    """run_stages with a tap must (a) reproduce forward() exactly and
+    (b) let a head-only invocation on the tap reproduce the full output —
+    the fx-free FedSTIL split (models/resnet.py:STAGES)."""
+    import torch
+    from flreid_amd.models.resnet import resnet18
+
+    torch.manual_seed(0)
+    m = resnet18(num_classes=8, neck="bnneck", last_stride=1)
+    m.eval()
+    x = torch.randn(2, 3, 64, 32)
+
+    full = m(x)
+    out, tap = m.run_stages(x, start=0, tap=4)     # tap = layer4 input
+    assert torch.allclose(out, full, atol=1e-6)
+    assert tap is not None and tap.dim() == 4
+
+    head_out, _ = m.run_stages(tap, start=4)
+    assert torch.allclose(head_out, full, atol=1e-6)
+
+
+def test_stage_of_mapping():
+    from flreid_amd.models.resnet import resnet18
+
+    m = resnet18(num_classes=8, neck="bnneck")
+    assert m.stage_of("base.conv1") == 0
+    assert m.stage_of("base.layer1.0.conv1") == 1
+    assert m.stage_of("base.layer4.1.bn2") == 4
+    assert m.stage_of("classifier") == len(m.STAGES) - 1
+    assert m.stage_of("bottleneck") == len(m.STAGES) - 1
