@@ -55,3 +55,21 @@ def test_autocast_cpu_is_noop():
     with autocast("cpu"):
         x = torch.randn(2, 2) @ torch.randn(2, 2)
     assert x.dtype == torch.float32
+
+
+def test_distill_kl_matches_manual():
+    """DistillKL = T²/B · KL(log_softmax(s/T) ‖ softmax(t/T))
+    (ref:criterions/kd_loss.py:10-27)."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(3)
+    kd = criterions["distill_kl"](T=4.0)
+    s = torch.randn(6, 10, requires_grad=True)
+    t = torch.randn(6, 10)
+    loss = kd(score=s, target=t, feature=None)
+    expected = F.kl_div(F.log_softmax(s / 4.0, dim=1),
+                        F.softmax(t / 4.0, dim=1),
+                        reduction="sum") * (4.0 ** 2) / s.shape[0]
+    assert torch.allclose(loss, expected, atol=1e-6)
+    loss.backward()
+    assert torch.isfinite(s.grad).all()
