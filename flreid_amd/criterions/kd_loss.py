@@ -7,7 +7,7 @@ from flreid_amd.modules.criterion import CriterionModule
 
 
 class DistillKL(CriterionModule):
-    def __init__(self, temperature: float = 4.0, **kwargs):
+    def __init__(self, temperature: float = 1.0, **kwargs):
         super().__init__()
         for n, p in kwargs.items():
             setattr(self, n, p)

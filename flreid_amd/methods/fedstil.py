@@ -67,6 +67,10 @@ class Model(ModelModule):
 
         self.ids = set()
         self.examplars: Dict[int, List] = {}
+        # physical layout of 4-D exemplar rows in the live store: [H, W, C]
+        # (GPU tap layout) when True, [C, H, W] otherwise.  Checkpoints are
+        # always canonical NCHW; examplar_tensors converts lazily.
+        self.examplars_nhwc = False
         self._egraphs: Dict = {}     # hipGraph cache for eval forwards
 
         # head = earliest stage containing an adaptive leaf
@@ -124,11 +128,22 @@ class Model(ModelModule):
     def m(self) -> int:
         return math.ceil(self.lambda_k / max(1, len(self.ids)))
 
-    def examplar_tensors(self, device):
+    def examplar_tensors(self, device, nhwc: Optional[bool] = None):
         """Stacked (data, pids, classes) of the exemplar store; data lives on
-        `device` (HBM-resident across rounds — 288 GB budget)."""
+        `device` (HBM-resident across rounds — 288 GB budget).  When `nhwc`
+        is given and differs from the store's physical layout (e.g. a ckpt
+        restored NCHW into a GPU run whose taps are [H, W, C] rows), the
+        store is converted once in place."""
         if not self.examplars:
             return None, None, None
+        if nhwc is not None and nhwc != self.examplars_nhwc:
+            perm = (0, 2, 3, 1) if nhwc else (0, 3, 1, 2)
+            for pid in list(self.examplars):
+                protos_t, classes_t = self.examplars[pid]
+                if protos_t.dim() == 4:
+                    self.examplars[pid] = (
+                        protos_t.permute(*perm).contiguous(), classes_t)
+            self.examplars_nhwc = nhwc
         datas, pids, classes = [], [], []
         for pid, entry in self.examplars.items():
             protos_t, classes_t = entry
@@ -249,6 +264,12 @@ class Model(ModelModule):
         class_sel = classes.index_select(0, chosen.view(-1)).view(P, -1)
         for r, p in enumerate(persons):
             self.examplars[int(p)] = (proto_sel[r], class_sel[r])
+        # selected rows are views into proto_loader.data — record its layout
+        # (any ckpt-restored NCHW entries were already converted when
+        # examplar_tensors built the rehearsal set for this task)
+        if torch.is_tensor(stacked):
+            self.examplars_nhwc = bool(getattr(proto_loader, "nhwc_stored",
+                                               False))
 
     def reduce_examplars(self) -> None:
         for k in list(self.examplars):
@@ -284,6 +305,7 @@ class Model(ModelModule):
         re-loaded the ENTIRE state dict for a 10-tensor dispatch —
         ref:methods/fedstil.py:535-547)."""
         live = {n: t for n, t in self.net.state_dict(keep_vars=True).items()}
+        rebound = False
         with torch.no_grad():
             for section in ("global_weight", "global_weight_atten",
                             "adaptive_weights", "adaptive_bias", "bn_params",
@@ -303,6 +325,11 @@ class Model(ModelModule):
                         target.data = p.detach().clone().to(dst.device, dst.dtype)
                         if hasattr(target, "grad"):
                             target.grad = None
+                        rebound = True
+        if rebound:
+            # captured eval hipGraphs hold the OLD storage pointers — replaying
+            # them after a rebind reads freed/stale weights
+            self._egraphs.clear()
 
     def composed_upload(self) -> Dict[str, torch.Tensor]:
         """{name.global_weight: atten⊙W_glob + W_adapt} — what the client
@@ -422,7 +449,8 @@ class Operator(BaseReIDOperator):
         pids = torch.cat(pids)
         classes = torch.cat(classes)
 
-        ex_data, ex_pids, ex_classes = model.examplar_tensors(device)
+        ex_data, ex_pids, ex_classes = model.examplar_tensors(
+            device, nhwc=on_gpu and taps.dim() == 4)
         if ex_data is not None:
             all_data = torch.cat([ex_data, taps])
             all_pids = torch.cat([ex_pids, pids])
@@ -594,9 +622,14 @@ class Client(BaseReIDClient):
     # {pid: [(proto, class), ...]} layout on CPU while the live store is
     # device-resident stacked tensors
     @staticmethod
-    def _examplars_to_ckpt(examplars: Dict) -> Dict:
+    def _examplars_to_ckpt(examplars: Dict, nhwc: bool = False) -> Dict:
+        """Checkpoints always store canonical NCHW rows, whatever physical
+        layout the live (possibly [H, W, C]-row) store uses — a GPU-written
+        ckpt loads correctly into a CPU run and vice versa."""
         out = {}
         for pid, (protos_t, classes_t) in examplars.items():
+            if nhwc and protos_t.dim() == 4:
+                protos_t = protos_t.permute(0, 3, 1, 2)
             out[pid] = [(protos_t[i].cpu(), int(classes_t[i]))
                         for i in range(protos_t.shape[0])]
         return out
@@ -622,6 +655,7 @@ class Client(BaseReIDClient):
         if self.state_exists(f"{model_name}_examplars"):
             self.model.examplars = self._examplars_from_ckpt(
                 self.load_state(f"{model_name}_examplars", {}))
+            self.model.examplars_nhwc = False    # ckpts are canonical NCHW
 
     def save_model(self, model_name: str) -> None:
         # gate BEFORE building the ckpt payload: the full-state clone and the
@@ -631,7 +665,9 @@ class Client(BaseReIDClient):
             return
         self.save_state(model_name, self.model.model_state(), True)
         self.save_state(f"{model_name}_examplars",
-                        self._examplars_to_ckpt(self.model.examplars), True)
+                        self._examplars_to_ckpt(self.model.examplars,
+                                                self.model.examplars_nhwc),
+                        True)
 
     def update_model(self, params_state: Dict) -> None:
         self.model.update_model(params_state)

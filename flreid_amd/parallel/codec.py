@@ -75,11 +75,17 @@ def sync_client_states(ctx, local_uploads: Dict[str, Any]) -> Dict[str, Any]:
     # 1. flatten locally
     metas: Dict[str, Any] = {}
     flats: Dict[str, torch.Tensor] = {}
+    local_has_nonfloat = False
     for cname in sorted(local_uploads.keys()):
         tensors: List[Tuple[str, torch.Tensor]] = []
         skeleton = _flatten(local_uploads[cname], cname, tensors)
+        # integer/bool tensors must not round-trip through the float wire
+        # dtype (precision loss above 2^24 in fp32, far sooner in bf16):
+        # such states take the object-gather fallback below
+        local_has_nonfloat = local_has_nonfloat or any(
+            not t.is_floating_point() for _n, t in tensors)
         wire = _wire_dtype()
-        if tensors:
+        if tensors and not local_has_nonfloat:
             flat = torch.cat([t.detach().reshape(-1).to(wire)
                               for _n, t in tensors]).to(device)
         else:
@@ -89,13 +95,15 @@ def sync_client_states(ctx, local_uploads: Dict[str, Any]) -> Dict[str, Any]:
 
     # 2. metadata hop (tiny): skeletons + per-client payload sizes
     per_client_sizes = {c: f.numel() for c, f in flats.items()}
-    all_metas = ctx.all_gather_object((metas, per_client_sizes))
+    all_metas = ctx.all_gather_object((metas, per_client_sizes,
+                                       local_has_nonfloat))
 
-    strides = {n for _m, sizes in all_metas for n in sizes.values()}
-    counts = [len(sizes) for _m, sizes in all_metas]
+    strides = {n for _m, sizes, _nf in all_metas for n in sizes.values()}
+    counts = [len(sizes) for _m, sizes, _nf in all_metas]
+    any_nonfloat = any(nf for _m, _s, nf in all_metas)
 
     # equal-stride fast path: every client's payload has the same numel
-    if len(strides) == 1 and any(counts):
+    if len(strides) == 1 and any(counts) and not any_nonfloat:
         stride = next(iter(strides))
         max_clients = max(counts)
         local = torch.zeros(max_clients * stride, dtype=_wire_dtype(),
@@ -105,7 +113,7 @@ def sync_client_states(ctx, local_uploads: Dict[str, Any]) -> Dict[str, Any]:
         gathered = ctx.all_gather_flat(local)          # [W, max_clients*stride]
 
         out: Dict[str, Any] = {}
-        for rank, (rank_metas, rank_sizes) in enumerate(all_metas):
+        for rank, (rank_metas, rank_sizes, _nf) in enumerate(all_metas):
             for i, cname in enumerate(sorted(rank_sizes.keys())):
                 flat = gathered[rank, i * stride:(i + 1) * stride]
                 out[cname] = _unflatten_state(rank_metas[cname], flat)

@@ -607,3 +607,38 @@ def test_tensor_codec_ragged_fallback(tmp_path):
     port = _free_port()
     mp.spawn(_worker_codec_ragged, args=(2, port, str(tmp_path)), nprocs=2,
              join=True)
+
+
+def _worker_codec_int(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.codec import sync_client_states
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        # equal strides across ranks, but an int64 tensor whose values exceed
+        # 2^24 — packing through the float wire would corrupt it
+        big = (1 << 40) + rank
+        state = {
+            "train_cnt": rank,
+            "ids": torch.tensor([big, big + 1], dtype=torch.int64),
+            "w": torch.full((4,), float(rank)),
+        }
+        merged = sync_client_states(ctx, {f"client-{rank}": state})
+        assert set(merged) == {"client-0", "client-1"}
+        for r in (0, 1):
+            ids = merged[f"client-{r}"]["ids"]
+            assert ids.dtype == torch.int64
+            assert ids.tolist() == [(1 << 40) + r, (1 << 40) + r + 1]
+            assert torch.allclose(merged[f"client-{r}"]["w"],
+                                  torch.full((4,), float(r)))
+    finally:
+        destroy_context()
+
+
+def test_tensor_codec_int_tensors_take_object_path(tmp_path):
+    """Integer tensors must never round-trip through the float wire dtype
+    (precision loss above 2^24): states containing them take the
+    object-gather fallback and come back bit-exact."""
+    port = _free_port()
+    mp.spawn(_worker_codec_int, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)

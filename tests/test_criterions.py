@@ -63,7 +63,7 @@ def test_distill_kl_matches_manual():
     import torch.nn.functional as F
 
     torch.manual_seed(3)
-    kd = criterions["distill_kl"](T=4.0)
+    kd = criterions["distill_kl"](temperature=4.0)
     s = torch.randn(6, 10, requires_grad=True)
     t = torch.randn(6, 10)
     loss = kd(score=s, target=t, feature=None)
@@ -73,3 +73,11 @@ def test_distill_kl_matches_manual():
     assert torch.allclose(loss, expected, atol=1e-6)
     loss.backward()
     assert torch.isfinite(s.grad).all()
+
+
+def test_distill_kl_default_temperature_matches_reference():
+    """ref:criterions/kd_loss.py:16 defaults temperature=1.0; a config
+    enabling distill_kl without an explicit temperature must get reference
+    semantics."""
+    from flreid_amd.criterions.kd_loss import DistillKL
+    assert DistillKL().temperature == 1.0

@@ -230,3 +230,61 @@ def test_personalized_dispatch_mixture_manual(tmp_path, monkeypatch):
     expected = sum(uploads[c]["incremental_sw"]["w"] * float(m)
                    for c, m in zip(names, w))
     assert torch.allclose(out["w"], expected, atol=1e-6)
+
+
+def _tiny_fedstil_model(num_classes=8):
+    from flreid_amd.methods.fedstil import Model
+    from flreid_amd.models.resnet import resnet18
+    net = resnet18(num_classes=num_classes, last_stride=1, neck="bnneck")
+    for p in net.parameters():
+        p.requires_grad = False
+    for p in net.classifier.parameters():
+        p.requires_grad = True
+    return Model(net, lambda_k=4, atten_default=0.9)
+
+
+def test_examplars_ckpt_canonical_nchw_roundtrip():
+    """Ckpts store canonical NCHW rows regardless of the live store's
+    physical layout; loading restores NCHW and examplar_tensors converts
+    lazily to the requested layout."""
+    from flreid_amd.methods.fedstil import Client
+
+    c, h, w = 3, 4, 2
+    nchw = torch.randn(2, c, h, w)
+    nhwc_store = {7: (nchw.permute(0, 2, 3, 1).contiguous(),
+                      torch.tensor([0, 1]))}
+    ck = Client._examplars_to_ckpt(nhwc_store, nhwc=True)
+    assert ck[7][0][0].shape == (c, h, w)            # canonical NCHW items
+    restored = Client._examplars_from_ckpt(ck)
+    assert torch.allclose(restored[7][0], nchw)
+
+    model = _tiny_fedstil_model()
+    model.examplars = restored
+    model.examplars_nhwc = False
+    data, pids, classes = model.examplar_tensors("cpu", nhwc=True)
+    assert data.shape == (2, h, w, c)                # converted on demand
+    assert torch.allclose(data.permute(0, 3, 1, 2), nchw)
+    assert model.examplars_nhwc is True
+    # and back
+    data2, _, _ = model.examplar_tensors("cpu", nhwc=False)
+    assert torch.allclose(data2, nchw)
+
+
+def test_egraphs_cleared_on_param_rebind():
+    """update_model that rebinds a parameter's storage (shape change — the
+    stacked-atten growth path) must invalidate cached eval hipGraphs: a
+    captured graph holds the OLD storage pointers."""
+    model = _tiny_fedstil_model()
+    model._egraphs[("tap_fwd", (1, 3, 8, 8))] = object()
+
+    # same-shape dispatch: in-place copy, cache stays
+    gw = {n: p.detach().clone() + 0.5
+          for n, p in model.model_state()["global_weight"].items()}
+    model.update_model({"global_weight": gw})
+    assert model._egraphs
+
+    # grown stack dim: rebind -> cache cleared
+    gw2 = {n: torch.cat([p.unsqueeze(-1), p.unsqueeze(-1)], dim=-1)
+           for n, p in gw.items()}
+    model.update_model({"global_weight": gw2})
+    assert not model._egraphs
