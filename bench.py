@@ -33,6 +33,9 @@ os.environ.setdefault("FLREID_DISABLE_CKPT", "1")
 os.environ.setdefault("FLREID_FAST_CONV", "1")
 os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 os.environ.setdefault("FLREID_GPU_AUGMENT", "1")
+# bf16 wire for the upload gather: halves xGMI traffic (the FedSTIL upload is
+# ~125 MB/client in fp32); the aggregation itself stays fp32 on-device
+os.environ.setdefault("FLREID_COMM_DTYPE", "bf16")
 
 
 def _seed_miopen_db():

@@ -24,6 +24,46 @@ from typing import Any, Dict, List, Tuple
 import torch
 
 _SENTINEL = "__flreid_tensor__"
+_SCALAR = "__flreid_scalar__"
+
+# steady-state schema cache: the full skeleton gather (an O(world_size)
+# object serialisation) runs only when some rank's upload STRUCTURE changes
+# (first contact, task switch growing a state, ...).  Per round, only a
+# fingerprint + the scalar leaves (train_cnt etc.) ride the object hop.
+_META_CACHE: Dict[str, Any] = {"fps": None, "parts": None}
+
+
+def reset_schema_cache() -> None:
+    _META_CACHE["fps"] = None
+    _META_CACHE["parts"] = None
+
+
+def _split_scalars(skeleton: Any, values: List[Any]) -> Any:
+    """Replace every non-tensor leaf with a numbered placeholder, collecting
+    the values — the structure half is cacheable across rounds, the values
+    half changes every round."""
+    if isinstance(skeleton, tuple) and len(skeleton) == 3 and skeleton[0] == _SENTINEL:
+        return skeleton
+    if isinstance(skeleton, dict):
+        return {k: _split_scalars(v, values) for k, v in skeleton.items()}
+    if isinstance(skeleton, (list, tuple)):
+        out = [_split_scalars(v, values) for v in skeleton]
+        return tuple(out) if isinstance(skeleton, tuple) else out
+    values.append(skeleton)
+    return (_SCALAR, len(values) - 1)
+
+
+def _subst_scalars(structure: Any, values: List[Any]) -> Any:
+    if isinstance(structure, tuple) and len(structure) == 2 and structure[0] == _SCALAR:
+        return values[structure[1]]
+    if isinstance(structure, tuple) and len(structure) == 3 and structure[0] == _SENTINEL:
+        return structure
+    if isinstance(structure, dict):
+        return {k: _subst_scalars(v, values) for k, v in structure.items()}
+    if isinstance(structure, (list, tuple)):
+        out = [_subst_scalars(v, values) for v in structure]
+        return tuple(out) if isinstance(structure, tuple) else out
+    return structure
 
 
 def _wire_dtype() -> torch.dtype:
@@ -93,10 +133,37 @@ def sync_client_states(ctx, local_uploads: Dict[str, Any]) -> Dict[str, Any]:
         metas[cname] = skeleton
         flats[cname] = flat
 
-    # 2. metadata hop (tiny): skeletons + per-client payload sizes
+    # 2. metadata hop: per round only a schema fingerprint + the scalar
+    # leaves are serialised; the full skeleton gather runs on schema change
+    import hashlib
+    import pickle
+
     per_client_sizes = {c: f.numel() for c, f in flats.items()}
-    all_metas = ctx.all_gather_object((metas, per_client_sizes,
-                                       local_has_nonfloat))
+    structures: Dict[str, Any] = {}
+    scalars: Dict[str, List[Any]] = {}
+    for cname, skeleton in metas.items():
+        vals: List[Any] = []
+        structures[cname] = _split_scalars(skeleton, vals)
+        scalars[cname] = vals
+    local_fp = hashlib.sha1(pickle.dumps(
+        (structures, per_client_sizes, local_has_nonfloat))).hexdigest()
+
+    small = ctx.all_gather_object((local_fp, scalars))
+    fps = [fp for fp, _sc in small]
+    if _META_CACHE["fps"] == fps:
+        all_parts = _META_CACHE["parts"]
+    else:
+        all_parts = ctx.all_gather_object(
+            (structures, per_client_sizes, local_has_nonfloat))
+        _META_CACHE["fps"] = fps
+        _META_CACHE["parts"] = all_parts
+
+    all_metas = []
+    for rank, (structures_r, sizes_r, nf_r) in enumerate(all_parts):
+        scal_r = small[rank][1]
+        metas_r = {c: _subst_scalars(s, scal_r[c])
+                   for c, s in structures_r.items()}
+        all_metas.append((metas_r, sizes_r, nf_r))
 
     strides = {n for _m, sizes, _nf in all_metas for n in sizes.values()}
     counts = [len(sizes) for _m, sizes, _nf in all_metas]

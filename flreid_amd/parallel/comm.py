@@ -178,3 +178,5 @@ def destroy_context() -> None:
     if dist.is_initialized():
         dist.destroy_process_group()
     _CTX = None
+    from flreid_amd.parallel import codec
+    codec.reset_schema_cache()

@@ -221,7 +221,8 @@ class ExperimentStage:
 
         log.record(f"comm.{curr_round}",
                    {"dispatch_bytes": comm_down, "upload_bytes": comm_up})
-        log.sync(self.ctx)
+        log.maybe_sync(self.ctx, curr_round,
+                       int(exp_config["exp_opts"]["comm_rounds"]))
         if phase_timers_enabled():
             print(f"[phases r{curr_round}] {dump_phases()}", flush=True)
 

@@ -92,3 +92,18 @@ class ExperimentLog:
                     self.apply_remote(pend)
             self.flush()
         self.pending.clear()
+
+    def maybe_sync(self, ctx, curr_round: int, total_rounds: int) -> None:
+        """Round-batched sync: the cross-rank object gather runs every
+        FLREID_LOG_SYNC_EVERY rounds (default 10) and on the final round,
+        not per round — the per-round O(world_size) serialisation would
+        otherwise sit on the 8-rank critical path.  Pending records
+        accumulate locally in between (the decision is a pure function of
+        (round, interval), so every rank takes the collective together)."""
+        if ctx is None or not ctx.is_distributed:
+            self.flush()
+            self.pending.clear()
+            return
+        interval = max(1, int(os.environ.get("FLREID_LOG_SYNC_EVERY", "10")))
+        if curr_round % interval == 0 or curr_round >= total_rounds:
+            self.sync(ctx)

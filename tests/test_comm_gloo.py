@@ -642,3 +642,127 @@ def test_tensor_codec_int_tensors_take_object_path(tmp_path):
     port = _free_port()
     mp.spawn(_worker_codec_int, args=(2, port, str(tmp_path)), nprocs=2,
              join=True)
+
+
+def _worker_fedstil_n(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        common, exp = _fedstil_cfg(tmpdir, f"dist{world}")
+        state = _run_fedstil_rounds(ctx, common, exp)
+        if rank == 0:
+            with open(os.path.join(tmpdir, f"fedstil_state_{world}.pkl"), "wb") as f:
+                pickle.dump(state, f)
+    finally:
+        destroy_context()
+
+
+@pytest.mark.timeout(900)
+def test_fedstil_4rank_equals_single(tmp_path):
+    """Pre-verification for the driver's 8-GPU SCALE run: the same FedSTIL
+    equivalence must hold at world_size=4 (4 clients / 4 ranks, partial
+    participation)."""
+    tmpdir = str(tmp_path)
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    from flreid_amd.parallel.comm import FedContext
+    common, exp = _fedstil_cfg(tmpdir, "single4")
+    single = _run_fedstil_rounds(FedContext(), common, exp)
+
+    port = _free_port()
+    mp.spawn(_worker_fedstil_n, args=(4, port, tmpdir), nprocs=4, join=True)
+    with open(os.path.join(tmpdir, "fedstil_state_4.pkl"), "rb") as f:
+        dist_state = pickle.load(f)
+    assert set(single) == set(dist_state)
+    for n in single:
+        assert torch.allclose(single[n], dist_state[n], atol=1e-5), n
+
+
+def _worker_codec_8(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    from flreid_amd.parallel.codec import sync_client_states
+    from flreid_amd.parallel.comm import destroy_context, init_context
+    ctx = init_context(device="cpu")
+    try:
+        # schema-cache accounting: wrap the object gather and count the FULL
+        # skeleton gathers (3-tuples) vs the tiny per-round hops
+        calls = {"full": 0, "small": 0}
+        inner = ctx.all_gather_object
+
+        def counting(obj):
+            if isinstance(obj, tuple) and len(obj) == 3:
+                calls["full"] += 1
+            else:
+                calls["small"] += 1
+            return inner(obj)
+
+        ctx.all_gather_object = counting
+        for r in range(3):
+            state = {
+                "train_cnt": 10 * r + rank,     # scalar changes per round
+                "incremental_sw": {"w": torch.full((3, 2), float(rank + r))},
+            }
+            merged = sync_client_states(ctx, {f"client-{rank}": state})
+            assert set(merged) == {f"client-{i}" for i in range(world)}
+            for i in range(world):
+                st = merged[f"client-{i}"]
+                assert st["train_cnt"] == 10 * r + i
+                assert torch.allclose(st["incremental_sw"]["w"],
+                                      torch.full((3, 2), float(i + r)))
+        # identical schema all 3 rounds -> exactly ONE full skeleton gather
+        assert calls["full"] == 1, calls
+        assert calls["small"] == 3, calls
+    finally:
+        destroy_context()
+
+
+@pytest.mark.timeout(900)
+def test_codec_8rank_schema_cache(tmp_path):
+    """world_size=8 codec round-trip with changing scalars: per-round
+    object serialisation is one fingerprint+scalars hop; the full skeleton
+    gather runs once (parallel/codec.py schema cache)."""
+    port = _free_port()
+    mp.spawn(_worker_codec_8, args=(8, port, str(tmp_path)), nprocs=8,
+             join=True)
+
+
+def _worker_bench_cpu(rank, world, port, tmpdir):
+    _dist_env(rank, world, port, tmpdir)
+    import sys
+    sys.argv = ["bench.py", "--cpu", "--gpus", str(world), "--steps", "1",
+                "--warmup", "1", "--model", "resnet18", "--img", "32x16",
+                "--batch", "8", "--ids", "4", "--imgs-per-id", "2",
+                "--num-classes", "64", "--lambda-k", "8"]
+    os.chdir(tmpdir)
+    out_path = os.path.join(tmpdir, "bench_out.txt")
+    import importlib
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, repo_root)
+    bench = importlib.import_module("bench")
+    if rank == 0:
+        with open(out_path, "w") as f:
+            stdout, sys.stdout = sys.stdout, f
+            try:
+                bench.main()
+            finally:
+                sys.stdout = stdout
+    else:
+        bench.main()
+
+
+@pytest.mark.timeout(900)
+def test_bench_8rank_cpu_path(tmp_path):
+    """The exact `bench.py --gpus 8` path the driver will run for
+    SCALE_rNN, exercised as 8 CPU/gloo processes: one client per rank,
+    upload gather at world 8, one JSON result line from rank 0."""
+    port = _free_port()
+    mp.spawn(_worker_bench_cpu, args=(8, port, str(tmp_path)), nprocs=8,
+             join=True)
+    lines = [ln for ln in open(os.path.join(str(tmp_path), "bench_out.txt"))
+             if ln.strip().startswith("{")]
+    assert lines, "no JSON result line from rank 0"
+    result = json.loads(lines[-1])
+    assert result["n_gpus"] == 8
+    assert result["config"]["parallelism"] == "fed-dp8 (1 client/GPU)"
+    assert result["value"] > 0
