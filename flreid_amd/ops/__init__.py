@@ -240,10 +240,83 @@ def importance_update(importance: Dict[str, torch.Tensor],
 # ---------------------------------------------------------------------------
 
 kl_distance = ref.kl_distance
-kd_loss = ref.kd_loss
 quadratic_penalty = ref.quadratic_penalty
 l1_drift_fused = ref.l1_drift_fused
 # l1_drift is defined below: fused HIP multi-tensor path on GPU
+
+
+class _KdLossFn(torch.autograd.Function):
+    """Fused temperature-softmax KL distillation (K7, kd.hip): one kernel
+    computes loss + dL/dz_student; the eager chain is 5 kernels plus the
+    autograd graph."""
+
+    @staticmethod
+    def forward(ctx, zs, zt, temperature, ext):
+        zs_c = zs.detach().float().contiguous()
+        zt_c = zt.detach().float().contiguous()
+        B, C = zs_c.shape
+        row_loss = torch.empty(B, device=zs.device, dtype=torch.float32)
+        grad = torch.empty_like(zs_c)
+        ext.kd_fwd(zs_c.data_ptr(), zt_c.data_ptr(), row_loss.data_ptr(),
+                   grad.data_ptr(), B, C, float(temperature), _stream())
+        ctx.save_for_backward(grad)
+        ctx.zs_dtype = zs.dtype
+        return row_loss.sum()
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (grad,) = ctx.saved_tensors
+        return (grad * grad_out).to(ctx.zs_dtype), None, None, None
+
+
+def kd_loss(logits_student: torch.Tensor, logits_teacher: torch.Tensor,
+            temperature: float = 4.0) -> torch.Tensor:
+    if logits_student.is_cuda and logits_student.dim() == 2:
+        ext = _ext_or_raise("kd_fwd")
+        if ext is not None:
+            return _KdLossFn.apply(logits_student, logits_teacher,
+                                   temperature, ext)
+    return ref.kd_loss(logits_student, logits_teacher, temperature)
+
+
+class _IcarlDistillFn(torch.autograd.Function):
+    """Fused iCaRL distillation step (K7, kd.hip): BOTH BCE-with-logits
+    losses (one-hot classification + sigmoid-teacher distillation on the
+    first P columns) and the combined gradient in ONE pass
+    (ref:methods/icarl.py:216-236)."""
+
+    @staticmethod
+    def forward(ctx, score, target, prev, ext):
+        score_c = score.detach().float().contiguous()
+        prev_c = prev.detach().float().contiguous()
+        B, C = score_c.shape
+        P = prev_c.shape[1]
+        row_loss = torch.empty(B, device=score.device, dtype=torch.float32)
+        grad = torch.empty_like(score_c)
+        ext.icarl_distill(score_c.data_ptr(),
+                          target.long().contiguous().data_ptr(),
+                          prev_c.data_ptr(), row_loss.data_ptr(),
+                          grad.data_ptr(), B, C, P, _stream())
+        ctx.save_for_backward(grad)
+        ctx.score_dtype = score.dtype
+        return row_loss.sum()
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (grad,) = ctx.saved_tensors
+        return (grad * grad_out).to(ctx.score_dtype), None, None, None
+
+
+def icarl_distill_loss(score: torch.Tensor, target: torch.Tensor,
+                       prev_logits: torch.Tensor) -> torch.Tensor:
+    """clf BCE(score, onehot(target)) + distill BCE(score[:, :P],
+    sigmoid(prev_logits)) — fused on GPU, eager reference elsewhere."""
+    if (score.is_cuda and score.dim() == 2
+            and prev_logits.shape[1] <= score.shape[1]):
+        ext = _ext_or_raise("icarl_distill")
+        if ext is not None:
+            return _IcarlDistillFn.apply(score, target, prev_logits, ext)
+    return ref.icarl_distill_loss(score, target, prev_logits)
 
 
 class _TripletHardFn(torch.autograd.Function):

@@ -55,6 +55,11 @@ extern "C" void flreid_bn_train_bwd(const void*, const void*, void*,
                                     const float*, float*, float*, float*,
                                     float*, int64_t, int, int, int,
                                     hipStream_t);
+extern "C" void flreid_kd_fwd(const float*, const float*, float*, float*,
+                              int64_t, int64_t, float, hipStream_t);
+extern "C" void flreid_icarl_distill(const float*, const int64_t*,
+                                     const float*, float*, float*, int64_t,
+                                     int64_t, int64_t, hipStream_t);
 extern "C" void flreid_drift_fwd(const int64_t*, const int*, float*, int,
                                  hipStream_t);
 extern "C" void flreid_drift_bwd(const int64_t*, const int*, const float*,
@@ -197,6 +202,25 @@ PYBIND11_MODULE(_flreid_hip, m) {
                                       (float*)dbeta, (float*)part_a,
                                       (float*)part_b, M, C, relu, dtype,
                                       as_stream(stream));
+        });
+
+  m.def("kd_fwd",
+        [](uintptr_t zs, uintptr_t zt, uintptr_t row_loss, uintptr_t grad,
+           int64_t B, int64_t C, float temperature, uintptr_t stream) {
+          flreid::flreid_kd_fwd((const float*)zs, (const float*)zt,
+                                (float*)row_loss, (float*)grad, B, C,
+                                temperature, as_stream(stream));
+        });
+
+  m.def("icarl_distill",
+        [](uintptr_t score, uintptr_t target, uintptr_t prev,
+           uintptr_t row_loss, uintptr_t grad, int64_t B, int64_t C,
+           int64_t P, uintptr_t stream) {
+          flreid::flreid_icarl_distill((const float*)score,
+                                       (const int64_t*)target,
+                                       (const float*)prev, (float*)row_loss,
+                                       (float*)grad, B, C, P,
+                                       as_stream(stream));
         });
 
   m.def("drift_fwd",

@@ -104,6 +104,21 @@ def triplet_loss(feature: torch.Tensor, target: torch.Tensor,
     return loss
 
 
+def icarl_distill_loss(score: torch.Tensor, target: torch.Tensor,
+                       prev_logits: torch.Tensor) -> torch.Tensor:
+    """iCaRL distillation step: classification BCE vs one-hot plus
+    distillation BCE vs sigmoid of the stored pre-update logits over the
+    first P (old-class) columns (ref:methods/icarl.py:216-236)."""
+    from flreid_amd.tools.utils import get_one_hot
+
+    p = prev_logits.shape[1]
+    clf = F.binary_cross_entropy_with_logits(
+        score.float(), get_one_hot(target, score.shape[1]))
+    distill = F.binary_cross_entropy_with_logits(
+        score[:, :p].float(), torch.sigmoid(prev_logits.float()))
+    return clf + distill
+
+
 def kd_loss(logits_student: torch.Tensor, logits_teacher: torch.Tensor,
             temperature: float = 4.0) -> torch.Tensor:
     """Temperature-scaled KL distillation (ref:criterions/kd_loss.py:10-27)."""

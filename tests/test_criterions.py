@@ -81,3 +81,21 @@ def test_distill_kl_default_temperature_matches_reference():
     semantics."""
     from flreid_amd.criterions.kd_loss import DistillKL
     assert DistillKL().temperature == 1.0
+
+
+def test_icarl_distill_reference_composition():
+    """ref.icarl_distill_loss == clf BCE + distill BCE computed by hand."""
+    import torch.nn.functional as F
+
+    from flreid_amd.ops import reference as ref
+    from flreid_amd.tools.utils import get_one_hot
+
+    torch.manual_seed(5)
+    score = torch.randn(8, 12)
+    target = torch.randint(0, 12, (8,))
+    prev = torch.randn(8, 5)
+    got = ref.icarl_distill_loss(score, target, prev)
+    expected = (F.binary_cross_entropy_with_logits(score, get_one_hot(target, 12))
+                + F.binary_cross_entropy_with_logits(score[:, :5],
+                                                     torch.sigmoid(prev)))
+    assert torch.allclose(got, expected, atol=1e-6)

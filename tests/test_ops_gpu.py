@@ -478,3 +478,38 @@ def test_adaptive_conv1x1_fused_route(monkeypatch):
     assert torch.allclose(x.grad.float(), x2.grad.float(), atol=0.1, rtol=5e-2)
     assert torch.allclose(conv.adaptive_weight.grad,
                           conv2.adaptive_weight.grad, atol=0.05, rtol=5e-2)
+
+
+def test_kd_loss_fused_matches_reference():
+    """K7 fused temperature-softmax KL (kd.hip) vs the fp32 eager
+    reference (ref:criterions/kd_loss.py:10-27), loss AND grad."""
+    torch.manual_seed(7)
+    for T in (1.0, 4.0):
+        zs = torch.randn(64, 8000, device="cuda", requires_grad=True)
+        zt = torch.randn(64, 8000, device="cuda")
+        loss = ops.kd_loss(zs, zt, T)
+        loss.backward()
+        zs_ref = zs.detach().clone().requires_grad_(True)
+        expected = ref.kd_loss(zs_ref, zt, T)
+        expected.backward()
+        assert torch.allclose(loss, expected, atol=1e-4, rtol=1e-5)
+        assert torch.allclose(zs.grad, zs_ref.grad, atol=1e-6)
+
+
+def test_icarl_distill_fused_matches_reference():
+    """K7 fused iCaRL distillation (both BCE-with-logits terms + combined
+    grad, kd.hip) vs the eager two-loss composition
+    (ref:methods/icarl.py:216-236)."""
+    torch.manual_seed(11)
+    B, C, P = 32, 96, 40
+    score = torch.randn(B, C, device="cuda", requires_grad=True)
+    target = torch.randint(0, C, (B,), device="cuda")
+    prev = torch.randn(B, P, device="cuda")
+    loss = ops.icarl_distill_loss(score, target, prev)
+    loss.backward()
+
+    score_ref = score.detach().clone().requires_grad_(True)
+    expected = ref.icarl_distill_loss(score_ref, target, prev)
+    expected.backward()
+    assert torch.allclose(loss, expected, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(score.grad, score_ref.grad, atol=1e-7)
