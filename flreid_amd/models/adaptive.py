@@ -129,6 +129,17 @@ class AdaptiveConv2d(AdaptiveBase):
 
     def forward(self, data: torch.Tensor) -> torch.Tensor:
         gw = self.global_weight
+        if (data.is_cuda and gw.dim() == 4 and gw.shape[2] == 3
+                and gw.shape[3] == 3 and _is_one(self.stride)
+                and _is_one(self.padding) and self.adaptive_bias is None
+                and not gw.requires_grad
+                and not self.global_weight_atten.requires_grad):
+            # hand-written K1 conv (conv3x3_img.hip): θ composed to bf16 in
+            # the weight fetch, halo-staged MFMA fwd + dgrad/wgrad kernels
+            y = ops.conv3x3_try(data, gw, self.global_weight_atten,
+                                self.adaptive_weight)
+            if y is not None:
+                return y
         if (data.is_cuda and gw.dim() == 4 and gw.shape[2] == 1
                 and gw.shape[3] == 1 and _is_one(self.stride)
                 and _is_zero(self.padding)):

@@ -75,8 +75,21 @@ class EvalFusedBatchNorm1d(nn.BatchNorm1d):
         return super().forward(x)
 
 
+class FusedConv3x3(nn.Conv2d):
+    """3×3 s1p1 conv that dispatches to the hand-written CDNA4 halo kernel
+    (ops/csrc/conv3x3_img.hip — K1) on the ReID layer-4 shapes, with the
+    library conv as fallback.  State-dict identical to nn.Conv2d."""
+
+    def forward(self, x):
+        if self.stride == (1, 1) and self.bias is None:
+            y = ops.conv3x3_try(x, self.weight)
+            if y is not None:
+                return y
+        return super().forward(x)
+
+
 def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+    return FusedConv3x3(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 
 def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:

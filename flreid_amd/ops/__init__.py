@@ -703,6 +703,130 @@ def bn_train_2d(x: torch.Tensor, bn,
                               bn.momentum, bn.eps, nbt, relu)
 
 
+def _cl(t: torch.Tensor) -> torch.Tensor:
+    return t if t.is_contiguous(memory_format=torch.channels_last) \
+        else t.contiguous(memory_format=torch.channels_last)
+
+
+def compose_theta_bf16(ext, gw: torch.Tensor, atten: Optional[torch.Tensor],
+                       aw: Optional[torch.Tensor]) -> torch.Tensor:
+    """θ = atten⊙gw + aw composed DIRECTLY to bf16 in the weight's own
+    physical layout (compose2 kernel) — one pass, no fp32 θ in HBM, no
+    autocast cast kernel.  atten broadcasts over the LOGICAL last dim; for
+    channels-last 4-D weights the physical index of that dim is
+    (i / C) % kw."""
+    gw_d = gw.detach()
+    if gw_d.dim() == 4:
+        assert gw_d.is_contiguous(memory_format=torch.channels_last)
+        inner, L = gw_d.shape[1], gw_d.shape[-1]
+    else:
+        assert gw_d.is_contiguous()
+        inner, L = 1, gw_d.shape[-1]
+    out = torch.empty_like(gw_d, dtype=torch.bfloat16)
+    ext.compose2(gw_d.data_ptr(),
+                 atten.detach().float().contiguous().data_ptr() if atten is not None else 0,
+                 aw.detach().data_ptr() if aw is not None else 0,
+                 out.data_ptr(), gw_d.numel(), L, inner, _dt(gw_d), _BF16,
+                 _stream())
+    return out
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    """Hand-written CDNA4 3×3 s1p1 conv (conv3x3_img.hip) with the FedSTIL
+    composition fused into the weight production (K1).
+
+    fwd:   θ_bf16 = compose2(gw, atten, aw) (or a cast of a plain weight);
+           y = conv3x3_img_fwd(x, θ) — one-image-per-block halo kernel.
+    dgrad: dx = conv3x3_img_fwd(dy, wflip(θ)) — same kernel, flipped
+           transposed taps.
+    wgrad: dθ = conv3x3_wgrad(dy, x), fp32 accumulate/output; for the
+           adaptive layer dθ IS d(adaptive_weight) (identity composition),
+           so no bf16 round-trip on the weight gradient.
+    Only frozen gw/atten are supported (FedSTIL trains aw alone;
+    plain nn.Conv2d passes the weight in slot 1)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, atten, aw):
+        ext = _ext_or_raise("conv3x3_img_fwd")
+        x_bf = _cl(x.detach().to(torch.bfloat16))
+        n, c, h, w = x_bf.shape
+        if aw is not None or atten is not None or weight.dtype != torch.bfloat16:
+            theta = compose_theta_bf16(ext, weight, atten, aw)
+        else:
+            theta = _cl(weight.detach())
+        k = weight.shape[0]
+        y = torch.empty(n, k, h, w, device=x.device, dtype=torch.bfloat16
+                        ).to(memory_format=torch.channels_last)
+        ext.conv3x3_img_fwd(x_bf.data_ptr(), theta.data_ptr(), y.data_ptr(),
+                            n, h, w, c, k, _stream())
+        ctx.save_for_backward(x_bf, theta)
+        ctx.x_dtype = x.dtype
+        ctx.w_dtype = weight.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _ext_or_raise("conv3x3_img_fwd")
+        x_bf, theta = ctx.saved_tensors
+        n, c, h, w = x_bf.shape
+        k = theta.shape[0]
+        dy_bf = _cl(dy.to(torch.bfloat16))
+        dx = None
+        if ctx.needs_input_grad[0]:
+            wt = torch.empty(c, k, 3, 3, device=dy.device, dtype=torch.bfloat16
+                             ).to(memory_format=torch.channels_last)
+            ext.conv3x3_wflip(theta.data_ptr(), wt.data_ptr(), c, k, _stream())
+            dx = torch.empty(n, c, h, w, device=dy.device, dtype=torch.bfloat16
+                             ).to(memory_format=torch.channels_last)
+            ext.conv3x3_img_fwd(dy_bf.data_ptr(), wt.data_ptr(), dx.data_ptr(),
+                                n, h, w, k, c, _stream())
+            if ctx.x_dtype != torch.bfloat16:
+                dx = dx.to(ctx.x_dtype)
+        d_weight = d_aw = None
+        if ctx.needs_input_grad[1] or ctx.needs_input_grad[3]:
+            dw = torch.empty(k, c, 3, 3, device=dy.device, dtype=torch.float32
+                             ).to(memory_format=torch.channels_last)
+            ext.conv3x3_wgrad(dy_bf.data_ptr(), x_bf.data_ptr(), dw.data_ptr(),
+                              n, h, w, c, k, _stream())
+            if ctx.needs_input_grad[3]:
+                d_aw = dw              # identity composition, fp32 direct
+            else:
+                d_weight = dw if ctx.w_dtype == torch.float32 else dw.to(ctx.w_dtype)
+        return dx, d_weight, None, d_aw
+
+
+def conv3x3_try(x: torch.Tensor, weight: torch.Tensor,
+                atten: Optional[torch.Tensor] = None,
+                aw: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
+    """Dispatch guard for the hand-written 3×3 s1p1 conv: returns None when
+    out of regime (caller falls back to the library conv).  Regime: the
+    ReID layer-4 shapes — H·W ≤ 128 (one image per block), H·W % 16 == 0,
+    C % 32, K % 32, channels-last weights, frozen gw/atten, bf16 compute
+    (native or autocast)."""
+    if not x.is_cuda or x.dim() != 4 or not extension_available():
+        return None
+    if os.environ.get("FLREID_NO_FUSED_CONV", "0") == "1":
+        return None
+    n, c, h, w = x.shape
+    k = weight.shape[0]
+    hw = h * w
+    if hw > 128 or hw % 16 or h + 2 > 18 or w + 2 > 10:
+        return None
+    if c % 32 or k % 32:
+        return None
+    if x.dtype != torch.bfloat16 and not torch.is_autocast_enabled():
+        return None
+    if weight.requires_grad and (aw is not None or atten is not None):
+        return None                      # composed path trains aw only
+    if atten is not None and atten.requires_grad:
+        return None
+    if weight.dim() != 4 or not weight.is_contiguous(memory_format=torch.channels_last):
+        return None
+    if aw is not None and not aw.is_contiguous(memory_format=torch.channels_last):
+        return None
+    return _Conv3x3Fn.apply(x, weight, atten, aw)
+
+
 def conv3x3_fwd_nhwc(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """Hand-written 3×3 s1 p1 NHWC bf16 conv forward (K1; frozen-backbone
     eval path).  x: channels-last bf16 [N, C, H, W]; weight: fp32

@@ -1,0 +1,429 @@
+// K1 production conv suite: 3×3 stride-1 pad-1 NHWC bf16, specialized for
+// the ReID layer-4 regime (H·W ≤ 128: one IMAGE per block).
+//
+// The ref hot path (ref:models/resnet.py:93-141, the _Bottleneck conv chain
+// at 16×8 spatial with last_stride=1) runs these shapes for every training
+// step of every method and for the frozen-backbone eval forwards.
+//
+// Design vs the generic nine-shifted-GEMM kernel (conv3x3.hip):
+//  - one image per block: the 3×3 taps re-read the SAME image rows, so the
+//    image's C-tile is staged ONCE into a zero-padded LDS halo
+//    [(H+2)·(W+2)][32] and all nine taps read LDS — 9× less global x traffic
+//    and NO border masking in the MFMA loop (the pad rows/cols are zero);
+//  - bf16 weights (the composed θ is produced in bf16 by the compose
+//    kernel) — half the weight bytes of the fp32-weight generic kernel;
+//  - the LDS c-dimension is stored PERMUTED (p(c) = ((c%16)>>2)*8 +
+//    (c>>4)*4 + (c&3)) so one MFMA fragment (the 8 bf16 a lane feeds to
+//    v_mfma_f32_16x16x32_bf16: k = kg+e and 16+kg+e') is 16 CONTIGUOUS
+//    bytes -> one ds_read_b128 per fragment instead of 8 scalar reads;
+//  - cell stride 40 elems (80 B) keeps b128 rows 16B-aligned and spreads
+//    banks;
+//  - T14 pipeline: per c-tile, one barrier; global loads for tile t+2 are
+//    issued before the MFMA burst on tile t (72 MFMAs/wave per barrier);
+//  - grid = (K/64, n_images): with K/64 == 8 the k-block equals the XCD id
+//    (blocks dispatch round-robin by linear id), so each XCD's L2 holds one
+//    295 KB weight slice instead of all 2.4 MB.
+//
+// dgrad = this same kernel applied to (dy, flipped-transposed θ): dx =
+// conv3x3_s1p1(dy, wT) with wT[c][r][s][k] = w[k][2-r][2-s][c]
+// (conv3x3_wflip below builds wT in one gather pass).
+//
+// wgrad (conv3x3_wgrad_kernel): per-tap M-reduction GEMM
+// dθ[k,c,(r,s)] = Σ_m dy[m,k]·x[shift_rs(m),c], fp32 accumulate/output,
+// border masking applied at the staging loads.
+
+#include "common.h"
+
+#include <algorithm>
+
+namespace flreid {
+
+using ibf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using if32x4 = __attribute__((ext_vector_type(4))) float;
+using u16x8 = __attribute__((ext_vector_type(8))) unsigned short;
+
+constexpr int CI_CS = 40;        // padded per-cell c-stride (elems)
+constexpr int CI_CELLS = 18 * 10;  // max (H+2)*(W+2)
+constexpr int CI_BK = 32;        // input-channel tile
+constexpr int CI_BN = 64;        // output channels per block
+
+// LDS c-permutation: fragment phys chunk base for c-chunk j (c = 8j..8j+7)
+__device__ __forceinline__ int ci_pbase(int j) {
+  // j=0 -> {0..3, 8..11}; j=1 -> {16..19, 24..27}; j=2 -> {4..7, 12..15};
+  // j=3 -> {20..23, 28..31}   (see p(c) above)
+  const int base[4] = {0, 16, 4, 20};
+  return base[j];
+}
+
+template <int MF>
+__global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
+    const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ W,
+    __hip_bfloat16* __restrict__ Y, int H, int Wd, int C, int K) {
+  __shared__ __hip_bfloat16 lx[2][CI_CELLS * CI_CS];
+  __shared__ __hip_bfloat16 lw[2][9 * CI_BN * CI_CS];
+
+  const int k0 = blockIdx.x * CI_BN;      // k-block first: XCD affinity
+  const int img = blockIdx.y;
+  const int HW = H * Wd;                  // == MF * 16
+  const int Wp = Wd + 2;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int fn = wave * 16 + (lane & 15); // this wave's output channel row
+  const int kg = (lane >> 4);             // k-group 0..3 (chunk index)
+
+  // ---- zero the x halo buffers once (pads stay zero forever) ----
+  {
+    const int total = (H + 2) * Wp * CI_CS;
+    for (int i = tid; i < total; i += 256) {
+      lx[0][i] = __float2bfloat16(0.f);
+      lx[1][i] = __float2bfloat16(0.f);
+    }
+  }
+
+  // ---- per-thread staging slots ----
+  // x: slot = cell*4 + j  (cell = output pixel, j = 16B c-chunk)
+  const int xs_slots = HW * 4;
+  int x_cell[2], x_j[2];
+  int64_t x_gaddr[2];
+  int x_laddr[2];
+  {
+    const int64_t img_base = (int64_t)img * HW * C;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int slot = tid + 256 * i;
+      const int cell = slot >> 2, j = slot & 3;
+      x_cell[i] = cell;
+      x_j[i] = j;
+      if (slot < xs_slots) {
+        const int h = cell / Wd, w = cell % Wd;
+        x_gaddr[i] = img_base + (int64_t)cell * C + j * 8;
+        x_laddr[i] = ((h + 1) * Wp + (w + 1)) * CI_CS + ci_pbase(j);
+      } else {
+        x_gaddr[i] = -1;
+        x_laddr[i] = 0;
+      }
+    }
+  }
+  // w: 9 slots, one per tap: k = tid>>2, j = tid&3
+  const int wk = tid >> 2, wj = tid & 3;
+  const bool w_ok = (k0 + wk) < K;
+  int64_t w_gbase = ((int64_t)(k0 + wk) * 9) * C + wj * 8;  // + tap*C + ct*32
+  const int w_lbase = wk * CI_CS + ci_pbase(wj);
+
+  const int NT = C / CI_BK;
+
+  u16x8 xr[2];
+  u16x8 wr[9];
+
+  auto load_tile = [&](int ct) {
+    const int64_t coff = (int64_t)ct * CI_BK;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      if (x_gaddr[i] >= 0) {
+        xr[i] = *(const u16x8*)(X + x_gaddr[i] + coff);
+      }
+    }
+#pragma unroll
+    for (int t = 0; t < 9; ++t) {
+      if (w_ok) {
+        wr[t] = *(const u16x8*)(W + w_gbase + (int64_t)t * C + coff);
+      }
+    }
+  };
+
+  auto store_tile = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      if (x_gaddr[i] >= 0) {
+        __hip_bfloat16* dst = &lx[buf][x_laddr[i]];
+        *(uint64_t*)dst = *(const uint64_t*)&xr[i];      // phys p..p+3
+        *(uint64_t*)(dst + 8) = *(((const uint64_t*)&xr[i]) + 1);  // p+8..p+11
+      }
+    }
+#pragma unroll
+    for (int t = 0; t < 9; ++t) {
+      __hip_bfloat16* dst = &lw[buf][t * CI_BN * CI_CS + w_lbase];
+      const uint64_t lo = w_ok ? *(const uint64_t*)&wr[t] : 0ull;
+      const uint64_t hi = w_ok ? *(((const uint64_t*)&wr[t]) + 1) : 0ull;
+      *(uint64_t*)dst = lo;
+      *(uint64_t*)(dst + 8) = hi;
+    }
+  };
+
+  // per-(thread, mf) halo cell offsets for the afrag reads
+  int mcell[MF];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    const int m = mf * 16 + (lane & 15);
+    const int h = m / Wd, w = m % Wd;
+    mcell[mf] = ((h + 1) * Wp + (w + 1)) * CI_CS + 8 * kg;
+  }
+
+  if32x4 acc[MF] = {};
+
+  load_tile(0);
+  __syncthreads();   // after the zero pass
+  store_tile(0);
+  if (NT > 1) load_tile(1);
+  __syncthreads();
+
+  for (int ct = 0; ct < NT; ++ct) {
+    if (ct + 1 < NT) {
+      store_tile((ct + 1) & 1);
+      if (ct + 2 < NT) load_tile(ct + 2);
+    }
+    const __hip_bfloat16* xb = lx[ct & 1];
+    const __hip_bfloat16* wb = lw[ct & 1];
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+#pragma unroll
+      for (int s = 0; s < 3; ++s) {
+        const int tap = r * 3 + s;
+        const ibf16x8 bfrag = *(const ibf16x8*)(
+            wb + tap * CI_BN * CI_CS + fn * CI_CS + 8 * kg);
+        const int toff = ((r - 1) * Wp + (s - 1)) * CI_CS;
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          const ibf16x8 afrag = *(const ibf16x8*)(xb + mcell[mf] + toff);
+          acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                            acc[mf], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  const int kc = k0 + fn;
+  if (kc >= K) return;
+  const int64_t out_base = (int64_t)img * HW * K + kc;
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int m = mf * 16 + (lane >> 4) * 4 + reg;
+      Y[out_base + (int64_t)m * K] = __float2bfloat16(acc[mf][reg]);
+    }
+  }
+}
+
+extern "C" void flreid_conv3x3_img_fwd(const void* X, const void* W, void* Y,
+                                       int NB, int H, int Wd, int C, int K,
+                                       hipStream_t stream) {
+  const int HW = H * Wd;
+  if (HW > 128 || (HW & 15) || C % CI_BK || K % 16 || H + 2 > 18 ||
+      Wd + 2 > 10) {
+    throw std::runtime_error("conv3x3_img_fwd: shape out of regime");
+  }
+  dim3 grid((K + CI_BN - 1) / CI_BN, NB);
+  const int mf = HW >> 4;
+  switch (mf) {
+#define FLREID_CI_CASE(MF)                                                   \
+  case MF:                                                                   \
+    hipLaunchKernelGGL(conv3x3_img_fwd_kernel<MF>, grid, dim3(256), 0,       \
+                       stream, (const __hip_bfloat16*)X,                     \
+                       (const __hip_bfloat16*)W, (__hip_bfloat16*)Y, H, Wd,  \
+                       C, K);                                                \
+    break;
+    FLREID_CI_CASE(1)
+    FLREID_CI_CASE(2)
+    FLREID_CI_CASE(3)
+    FLREID_CI_CASE(4)
+    FLREID_CI_CASE(5)
+    FLREID_CI_CASE(6)
+    FLREID_CI_CASE(7)
+    FLREID_CI_CASE(8)
+#undef FLREID_CI_CASE
+    default:
+      throw std::runtime_error("conv3x3_img_fwd: bad MF");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// weight transpose-flip for dgrad: wT[c][r][s][k] = w[k][2-r][2-s][c]
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void conv3x3_wflip_kernel(
+    const __hip_bfloat16* __restrict__ W, __hip_bfloat16* __restrict__ WT,
+    int C, int K) {
+  // one thread per output element, writes coalesced over k
+  const int64_t total = (int64_t)C * 9 * K;
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * 256) {
+    const int k = (int)(i % K);
+    const int64_t rest = i / K;
+    const int tap = (int)(rest % 9);
+    const int c = (int)(rest / 9);
+    const int r = tap / 3, s = tap % 3;
+    WT[i] = W[((int64_t)k * 9 + (2 - r) * 3 + (2 - s)) * C + c];
+  }
+}
+
+extern "C" void flreid_conv3x3_wflip(const void* W, void* WT, int C, int K,
+                                     hipStream_t stream) {
+  const int64_t total = (int64_t)C * 9 * K;
+  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 4096);
+  hipLaunchKernelGGL(conv3x3_wflip_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const __hip_bfloat16*)W, (__hip_bfloat16*)WT, C, K);
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// wgrad: dθ[k][r][s][c] = Σ_m dy[m][k] · x[shift_rs(m)][c]
+// ---------------------------------------------------------------------------
+// Per-tap M-reduction GEMM: A = dyᵀ (k rows), B = xᵀ (c rows), reduction
+// over flattened NHWC rows m in 32-row chunks.  Both tiles staged
+// TRANSPOSED into LDS (rows = out dims, cols = m, permuted for b128
+// fragments).  The tap's x-shift and border mask are applied at the
+// staging loads.  Output fp32, each block owns its (k-tile, c-tile, tap)
+// exclusively — no atomics.
+
+constexpr int WG_BK = 64;    // k rows per block
+constexpr int WG_BC = 128;   // c cols per block
+constexpr int WG_BM = 32;    // m chunk
+
+__device__ __forceinline__ int wg_perm(int m) {
+  return ((m & 15) >> 2) * 8 + (m >> 4) * 4 + (m & 3);
+}
+
+__global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ DY, const __hip_bfloat16* __restrict__ X,
+    float* __restrict__ DW, int NB, int H, int Wd, int C, int K) {
+  __shared__ __hip_bfloat16 ldy[2][WG_BK * CI_CS];
+  __shared__ __hip_bfloat16 lxc[2][WG_BC * CI_CS];
+
+  const int kb = blockIdx.x;          // k-block first: XCD dy-affinity
+  const int cb = blockIdx.y;
+  const int tap = blockIdx.z;
+  const int r = tap / 3 - 1, s = tap % 3 - 1;
+  const int k0 = kb * WG_BK, c0 = cb * WG_BC;
+  const int HW = H * Wd;
+  const int64_t M = (int64_t)NB * HW;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wk = wave & 1, wc = wave >> 1;   // 2(k) × 2(c) wave grid
+  const int kg = lane >> 4;
+
+  // staging slots --------------------------------------------------------
+  // dy: 32m × 64k in 16B units = 256 slots: m_l = tid>>3, j = tid&7
+  const int dy_m = tid >> 3, dy_j = tid & 7;
+  // x: 32m × 128c = 512 slots: 2 per thread
+  const int x_m[2] = {tid >> 4, (tid + 256) >> 4};
+  const int x_j[2] = {tid & 15, (tid + 256) & 15};
+
+  u16x8 dyr, xr[2];
+  bool dy_v = false, x_v[2] = {false, false};
+
+  auto load_tile = [&](int64_t m0) {
+    {
+      const int64_t m = m0 + dy_m;
+      dy_v = m < M && (k0 + dy_j * 8) < K;
+      dyr = dy_v ? *(const u16x8*)(DY + m * K + k0 + dy_j * 8) : u16x8{};
+    }
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int64_t m = m0 + x_m[i];
+      bool ok = m < M;
+      int64_t src = 0;
+      if (ok) {
+        const int n = (int)(m / HW);
+        const int rem = (int)(m % HW);
+        const int h = rem / Wd + r, w = rem % Wd + s;
+        ok = h >= 0 && h < H && w >= 0 && w < Wd;
+        src = ((int64_t)n * HW + h * Wd + w) * C + c0 + x_j[i] * 8;
+        ok = ok && (c0 + x_j[i] * 8) < C;
+      }
+      x_v[i] = ok;
+      xr[i] = ok ? *(const u16x8*)(X + src) : u16x8{};
+    }
+  };
+
+  auto store_tile = [&](int buf) {
+    {
+      // transposed: 8 k-rows, column p(m)
+      const int pm = wg_perm(dy_m);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        ldy[buf][(dy_j * 8 + e) * CI_CS + pm] =
+            dy_v ? *((const __hip_bfloat16*)&dyr + e) : __float2bfloat16(0.f);
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int pm = wg_perm(x_m[i]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        lxc[buf][(x_j[i] * 8 + e) * CI_CS + pm] =
+            x_v[i] ? *((const __hip_bfloat16*)&xr[i] + e) : __float2bfloat16(0.f);
+      }
+    }
+  };
+
+  if32x4 acc[2][4] = {};
+
+  const int64_t NIT = (M + WG_BM - 1) / WG_BM;
+  load_tile(0);
+  store_tile(0);
+  if (NIT > 1) load_tile(WG_BM);
+  __syncthreads();
+
+  for (int64_t it = 0; it < NIT; ++it) {
+    if (it + 1 < NIT) {
+      store_tile((int)((it + 1) & 1));
+      if (it + 2 < NIT) load_tile((it + 2) * WG_BM);
+    }
+    const __hip_bfloat16* db = ldy[it & 1];
+    const __hip_bfloat16* xb = lxc[it & 1];
+    ibf16x8 afrag[2];
+#pragma unroll
+    for (int fk = 0; fk < 2; ++fk) {
+      afrag[fk] = *(const ibf16x8*)(
+          db + (wk * 32 + fk * 16 + (lane & 15)) * CI_CS + 8 * kg);
+    }
+#pragma unroll
+    for (int fc = 0; fc < 4; ++fc) {
+      const ibf16x8 bfrag = *(const ibf16x8*)(
+          xb + (wc * 64 + fc * 16 + (lane & 15)) * CI_CS + 8 * kg);
+#pragma unroll
+      for (int fk = 0; fk < 2; ++fk) {
+        acc[fk][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[fk], bfrag, acc[fk][fc], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // D rows = k (A), cols = c (B)
+#pragma unroll
+  for (int fk = 0; fk < 2; ++fk) {
+#pragma unroll
+    for (int fc = 0; fc < 4; ++fc) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int k = k0 + wk * 32 + fk * 16 + (lane >> 4) * 4 + reg;
+        const int c = c0 + wc * 64 + fc * 16 + (lane & 15);
+        if (k < K && c < C) {
+          DW[((int64_t)k * 9 + tap) * C + c] = acc[fk][fc][reg];
+        }
+      }
+    }
+  }
+}
+
+extern "C" void flreid_conv3x3_wgrad(const void* DY, const void* X, float* DW,
+                                     int NB, int H, int Wd, int C, int K,
+                                     hipStream_t stream) {
+  if (C % 8 || K % 8) {
+    throw std::runtime_error("conv3x3_wgrad: C%8 or K%8 != 0");
+  }
+  dim3 grid((K + WG_BK - 1) / WG_BK, (C + WG_BC - 1) / WG_BC, 9);
+  hipLaunchKernelGGL(conv3x3_wgrad_kernel, grid, dim3(256), 0, stream,
+                     (const __hip_bfloat16*)DY, (const __hip_bfloat16*)X, DW,
+                     NB, H, Wd, C, K);
+  HIP_CHECK(hipGetLastError());
+}
+
+}  // namespace flreid

@@ -146,6 +146,51 @@ __global__ void compose_kernel(const T* __restrict__ gw,
                    fmaf(a, load_as_float(gw, i), load_as_float(aw, i)));
 }
 
+// compose2: layout-aware + mixed-dtype composition.  atten broadcasts over
+// the LOGICAL last weight dim; for channels-last conv weights the physical
+// index of that dim is (i / inner) % L (inner = C for [K][kh][kw][C]
+// storage, 1 for linear weights).  Tout=bf16 with Tin=f32 produces the
+// bf16 θ the MFMA conv/GEMM kernels consume directly — no autocast cast
+// pass, θ never exists in fp32 HBM.
+template <typename Tin, typename Tout>
+__global__ void compose2_kernel(const Tin* __restrict__ gw,
+                                const float* __restrict__ atten,
+                                const Tin* __restrict__ aw,
+                                Tout* __restrict__ out, int64_t numel,
+                                int64_t L, int64_t inner) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= numel) return;
+  const float a = atten ? atten[(i / inner) % L] : 1.0f;
+  const float v = fmaf(a, load_as_float(gw, i),
+                       aw ? load_as_float(aw, i) : 0.0f);
+  store_from_float(out, i, v);
+}
+
+extern "C" void flreid_compose2(const void* gw, const float* atten,
+                                const void* aw, void* out, int64_t numel,
+                                int64_t L, int64_t inner, int in_dtype,
+                                int out_dtype, hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  dim3 grid((unsigned)((numel + BLOCK - 1) / BLOCK)), block(BLOCK);
+  if (in_dtype == kF32 && out_dtype == kBF16) {
+    hipLaunchKernelGGL((compose2_kernel<float, __hip_bfloat16>), grid, block,
+                       0, stream, (const float*)gw, atten, (const float*)aw,
+                       (__hip_bfloat16*)out, numel, L, inner);
+  } else if (in_dtype == kF32 && out_dtype == kF32) {
+    hipLaunchKernelGGL((compose2_kernel<float, float>), grid, block, 0,
+                       stream, (const float*)gw, atten, (const float*)aw,
+                       (float*)out, numel, L, inner);
+  } else if (in_dtype == kBF16 && out_dtype == kBF16) {
+    hipLaunchKernelGGL((compose2_kernel<__hip_bfloat16, __hip_bfloat16>),
+                       grid, block, 0, stream, (const __hip_bfloat16*)gw,
+                       atten, (const __hip_bfloat16*)aw,
+                       (__hip_bfloat16*)out, numel, L, inner);
+  } else {
+    throw std::runtime_error("compose2: unsupported dtype combination");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
 extern "C" void flreid_compose(const void* gw, const float* atten,
                                const void* aw, void* out, int64_t numel,
                                int64_t L, int dtype, hipStream_t stream) {

@@ -55,6 +55,15 @@ extern "C" void flreid_bn_train_bwd(const void*, const void*, void*,
                                     const float*, float*, float*, float*,
                                     float*, int64_t, int, int, int,
                                     hipStream_t);
+extern "C" void flreid_compose2(const void*, const float*, const void*,
+                                void*, int64_t, int64_t, int64_t, int, int,
+                                hipStream_t);
+extern "C" void flreid_conv3x3_img_fwd(const void*, const void*, void*, int,
+                                       int, int, int, int, hipStream_t);
+extern "C" void flreid_conv3x3_wflip(const void*, void*, int, int,
+                                     hipStream_t);
+extern "C" void flreid_conv3x3_wgrad(const void*, const void*, float*, int,
+                                     int, int, int, int, hipStream_t);
 extern "C" void flreid_kd_fwd(const float*, const float*, float*, float*,
                               int64_t, int64_t, float, hipStream_t);
 extern "C" void flreid_icarl_distill(const float*, const int64_t*,
@@ -202,6 +211,38 @@ PYBIND11_MODULE(_flreid_hip, m) {
                                       (float*)dbeta, (float*)part_a,
                                       (float*)part_b, M, C, relu, dtype,
                                       as_stream(stream));
+        });
+
+  m.def("compose2",
+        [](uintptr_t gw, uintptr_t atten, uintptr_t aw, uintptr_t out,
+           int64_t numel, int64_t L, int64_t inner, int in_dtype,
+           int out_dtype, uintptr_t stream) {
+          flreid::flreid_compose2((const void*)gw, (const float*)atten,
+                                  (const void*)aw, (void*)out, numel, L,
+                                  inner, in_dtype, out_dtype,
+                                  as_stream(stream));
+        });
+
+  m.def("conv3x3_img_fwd",
+        [](uintptr_t x, uintptr_t w, uintptr_t y, int NB, int H, int Wd,
+           int C, int K, uintptr_t stream) {
+          flreid::flreid_conv3x3_img_fwd((const void*)x, (const void*)w,
+                                         (void*)y, NB, H, Wd, C, K,
+                                         as_stream(stream));
+        });
+
+  m.def("conv3x3_wflip",
+        [](uintptr_t w, uintptr_t wt, int C, int K, uintptr_t stream) {
+          flreid::flreid_conv3x3_wflip((const void*)w, (void*)wt, C, K,
+                                       as_stream(stream));
+        });
+
+  m.def("conv3x3_wgrad",
+        [](uintptr_t dy, uintptr_t x, uintptr_t dw, int NB, int H, int Wd,
+           int C, int K, uintptr_t stream) {
+          flreid::flreid_conv3x3_wgrad((const void*)dy, (const void*)x,
+                                       (float*)dw, NB, H, Wd, C, K,
+                                       as_stream(stream));
         });
 
   m.def("kd_fwd",

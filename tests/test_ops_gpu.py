@@ -513,3 +513,98 @@ def test_icarl_distill_fused_matches_reference():
     expected.backward()
     assert torch.allclose(loss, expected, atol=1e-5, rtol=1e-5)
     assert torch.allclose(score.grad, score_ref.grad, atol=1e-7)
+
+
+def _cl4(t):
+    return t.to(memory_format=torch.channels_last)
+
+
+def test_conv3x3_img_fwd_matches_torch():
+    """K1 specialized one-image-per-block halo kernel (conv3x3_img.hip)
+    fwd vs the fp32 library conv on bf16-rounded inputs."""
+    torch.manual_seed(1)
+    for (n, c, h, w, k) in ((4, 32, 16, 8, 64), (3, 64, 8, 4, 32),
+                            (8, 512, 16, 8, 512), (2, 256, 16, 8, 512)):
+        x = _cl4(torch.randn(n, c, h, w, device="cuda").bfloat16())
+        wt = _cl4(torch.randn(k, c, 3, 3, device="cuda") * (1.0 / c))
+        out = ops.conv3x3_try(x, wt)
+        assert out is not None, (n, c, h, w, k)
+        assert out.dtype == torch.bfloat16
+        expected = torch.nn.functional.conv2d(
+            x.float(), wt.float().to(memory_format=torch.contiguous_format),
+            padding=1)
+        assert torch.allclose(out.float(), expected, atol=0.1, rtol=5e-2), \
+            (n, c, h, w, k, (out.float() - expected).abs().max())
+
+
+def test_conv3x3_img_plain_bwd_matches_torch():
+    """dgrad (flip-transpose + the same fwd kernel) and wgrad (per-tap
+    M-reduction MFMA kernel, fp32 out) vs torch autograd in fp32."""
+    torch.manual_seed(2)
+    n, c, h, w, k = 4, 64, 16, 8, 64
+    x = _cl4(torch.randn(n, c, h, w, device="cuda").bfloat16())
+    wt = _cl4(torch.randn(k, c, 3, 3, device="cuda") * (1.0 / c))
+    x1 = x.clone().requires_grad_(True)
+    w1 = wt.clone().requires_grad_(True)
+    out = ops.conv3x3_try(x1, w1)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    x2 = x.float().clone().requires_grad_(True)
+    w2 = wt.clone().requires_grad_(True)
+    ref_out = torch.nn.functional.conv2d(x2, w2, padding=1)
+    ref_out.backward(g.float())
+
+    assert w1.grad.dtype == torch.float32
+    # wgrad reduces over 8192 rows of bf16 products in fp32
+    assert torch.allclose(w1.grad, w2.grad, atol=0.5, rtol=5e-2), \
+        (w1.grad - w2.grad).abs().max()
+    assert torch.allclose(x1.grad.float(), x2.grad, atol=0.15, rtol=5e-2), \
+        (x1.grad.float() - x2.grad).abs().max()
+
+
+def test_conv3x3_img_adaptive_fwd_bwd():
+    """Composed path: θ = atten⊙gw + aw fused into the conv weight fetch;
+    d(aw) comes straight out of wgrad in fp32 (identity composition)."""
+    torch.manual_seed(3)
+    n, c, h, w, k = 2, 32, 16, 8, 32
+    x = _cl4(torch.randn(n, c, h, w, device="cuda").bfloat16())
+    gw = _cl4(torch.randn(k, c, 3, 3, device="cuda") * (1.0 / c))
+    atten = torch.full((3,), 0.9, device="cuda")
+    aw = _cl4((0.1 * gw).clone()).requires_grad_(True)
+    out = ops.conv3x3_try(x, gw, atten, aw)
+    assert out is not None
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    aw2 = aw.detach().clone().requires_grad_(True)
+    theta = atten.view(1, 1, 1, 3) * gw + aw2
+    ref = torch.nn.functional.conv2d(x.float(), theta, padding=1)
+    ref.backward(g.float())
+    assert torch.allclose(out.float(), ref.detach(), atol=0.1, rtol=5e-2)
+    assert torch.allclose(aw.grad, aw2.grad, atol=0.3, rtol=5e-2), \
+        (aw.grad - aw2.grad).abs().max()
+
+
+def test_conv3x3_try_regime_guard():
+    """Out-of-regime shapes must decline (caller falls back to library)."""
+    x = _cl4(torch.randn(2, 64, 32, 16, device="cuda").bfloat16())  # HW=512
+    wt = _cl4(torch.randn(64, 64, 3, 3, device="cuda"))
+    assert ops.conv3x3_try(x, wt) is None
+    x2 = _cl4(torch.randn(2, 24, 16, 8, device="cuda").bfloat16())  # C%32
+    wt2 = _cl4(torch.randn(64, 24, 3, 3, device="cuda"))
+    assert ops.conv3x3_try(x2, wt2) is None
+
+
+def test_compose2_bf16_channels_last():
+    """compose2: bf16-out composition in channels-last layout with the
+    atten broadcast over the LOGICAL last dim (kernel width)."""
+    from flreid_amd.ops import compose_theta_bf16, _load_extension
+    ext = _load_extension()
+    gw = _cl4(torch.randn(16, 32, 3, 3, device="cuda"))
+    aw = _cl4(torch.randn(16, 32, 3, 3, device="cuda"))
+    atten = torch.tensor([0.25, 0.5, 0.75], device="cuda")
+    out = compose_theta_bf16(ext, gw, atten, aw)
+    assert out.dtype == torch.bfloat16
+    expected = (atten.view(1, 1, 1, 3) * gw + aw).bfloat16()
+    assert torch.allclose(out.float(), expected.float(), atol=2e-2)
