@@ -40,8 +40,8 @@ def run_shape(n, c, h, w, k):
     ext = ops._load_extension()
 
     def ours_fwd():
-        y = torch.empty(n, k, h, w, device="cuda", dtype=torch.bfloat16) \
-            .to(memory_format=torch.channels_last)
+        y = torch.empty(n, k, h, w, device="cuda", dtype=torch.bfloat16,
+                        memory_format=torch.channels_last)
         ext.conv3x3_img_fwd(x.data_ptr(), wt_bf.data_ptr(), y.data_ptr(),
                             n, h, w, c, k, torch.cuda.current_stream().cuda_stream)
         return y
@@ -50,19 +50,19 @@ def run_shape(n, c, h, w, k):
         return torch.nn.functional.conv2d(x, wt_bf, padding=1)
 
     def ours_dgrad():
-        wtt = torch.empty(c, k, 3, 3, device="cuda", dtype=torch.bfloat16) \
-            .to(memory_format=torch.channels_last)
+        wtt = torch.empty(c, k, 3, 3, device="cuda", dtype=torch.bfloat16,
+                          memory_format=torch.channels_last)
         ext.conv3x3_wflip(wt_bf.data_ptr(), wtt.data_ptr(), c, k,
                           torch.cuda.current_stream().cuda_stream)
-        dx = torch.empty(n, c, h, w, device="cuda", dtype=torch.bfloat16) \
-            .to(memory_format=torch.channels_last)
+        dx = torch.empty(n, c, h, w, device="cuda", dtype=torch.bfloat16,
+                         memory_format=torch.channels_last)
         ext.conv3x3_img_fwd(dy.data_ptr(), wtt.data_ptr(), dx.data_ptr(),
                             n, h, w, k, c, torch.cuda.current_stream().cuda_stream)
         return dx
 
     def ours_wgrad():
-        dw = torch.empty(k, c, 3, 3, device="cuda", dtype=torch.float32) \
-            .to(memory_format=torch.channels_last)
+        dw = torch.zeros(k, c, 3, 3, device="cuda", dtype=torch.float32,
+                         memory_format=torch.channels_last)
         ext.conv3x3_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr(),
                           n, h, w, c, k, torch.cuda.current_stream().cuda_stream)
         return dw

@@ -167,11 +167,16 @@ class PatchMerging(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         h, w = self.input_resolution
         b, L, c = x.shape
-        x = x.view(b, h, w, c)
-        x = torch.cat([x[:, 0::2, 0::2], x[:, 1::2, 0::2],
-                       x[:, 0::2, 1::2], x[:, 1::2, 1::2]], dim=-1)
-        x = x.view(b, -1, 4 * c)
-        return self.reduction(self.norm(x))
+        # fused K4 path (patch_merge.hip): gather + LN in one kernel, the
+        # concat tensor never materialises; eager fallback elsewhere
+        y = ops.patch_merge_ln(x, self.norm.weight, self.norm.bias, h, w,
+                               self.norm.eps)
+        if y is None:
+            x = x.view(b, h, w, c)
+            x = torch.cat([x[:, 0::2, 0::2], x[:, 1::2, 0::2],
+                           x[:, 0::2, 1::2], x[:, 1::2, 1::2]], dim=-1)
+            y = self.norm(x.view(b, -1, 4 * c))
+        return self.reduction(y)
 
 
 class SwinStage(nn.Module):

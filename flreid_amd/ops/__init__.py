@@ -396,30 +396,139 @@ def cmc_map(query_features, query_labels, gallery_features, gallery_labels,
     return total_cmc / q, total_ap / q
 
 
-def window_attention(q, k, v, bias, mask, scale, dropout=None):
-    """Swin window MHSA (K3): fused HIP kernel for the no-grad GPU path (the
-    frozen-backbone prototype capture + eval forward — the FedSTIL-Swin hot
-    loop); eager composition where autograd or dropout is needed."""
-    if (q.is_cuda and not torch.is_grad_enabled() and dropout is None
-            and q.shape[-2] <= 64 and q.shape[-1] <= 64):
+def _window_attn_fwd_raw(ext, q, k, v, bias, mask, scale):
+    bw, h, n, d = q.shape
+    qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+    bias_c = bias.detach().contiguous().float()
+    if mask is not None:
+        mask_c = mask.detach().contiguous().float()
+        nw = mask_c.shape[0]
+        mask_ptr = mask_c.data_ptr()
+    else:
+        mask_c, nw, mask_ptr = None, 1, 0
+    out = torch.empty_like(qc)
+    ext.window_attn_fwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                        bias_c.data_ptr(), mask_ptr, out.data_ptr(),
+                        bw, h, n, d, nw, float(scale), _dt(qc), _stream())
+    return out, qc, kc, vc, bias_c, mask_c
+
+
+class _WindowAttnFn(torch.autograd.Function):
+    """Fused Swin window MHSA fwd+bwd (K3, window_attn.hip) — the TRAINING
+    path: the backward recomputes S→P per (window, head) in LDS and emits
+    dQ/dK/dV plus dS; the relative-position-bias gradient is the window
+    reduction of dS (the bias-table gather backward stays in autograd)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, bias, mask, scale):
         ext = _ext_or_raise("window_attn_fwd")
-        if ext is not None:
-            bw, h, n, d = q.shape
-            qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
-            bias_c = bias.detach().contiguous().float()
-            if mask is not None:
-                mask_c = mask.detach().contiguous().float()
-                nw = mask_c.shape[0]
-                mask_ptr = mask_c.data_ptr()
-            else:
-                mask_c, nw, mask_ptr = None, 1, 0
-            out = torch.empty_like(qc)
-            ext.window_attn_fwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                                bias_c.data_ptr(), mask_ptr, out.data_ptr(),
-                                bw, h, n, d, nw, float(scale), _dt(qc),
-                                _stream())
+        out, qc, kc, vc, bias_c, mask_c = _window_attn_fwd_raw(
+            ext, q.detach(), k.detach(), v.detach(), bias, mask, scale)
+        ctx.save_for_backward(qc, kc, vc, bias_c)
+        ctx.mask_c = mask_c
+        ctx.scale = float(scale)
+        ctx.bias_requires = bias.requires_grad
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        ext = _ext_or_raise("window_attn_bwd")
+        qc, kc, vc, bias_c = ctx.saved_tensors
+        mask_c = ctx.mask_c
+        bw, h, n, d = qc.shape
+        nw = mask_c.shape[0] if mask_c is not None else 1
+        go = grad_out.contiguous().to(qc.dtype)
+        dq = torch.empty_like(qc)
+        dk = torch.empty_like(kc)
+        dv = torch.empty_like(vc)
+        ds = torch.empty(bw, h, n, n, device=qc.device, dtype=torch.float32)
+        ext.window_attn_bwd(qc.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                            bias_c.data_ptr(),
+                            mask_c.data_ptr() if mask_c is not None else 0,
+                            go.data_ptr(), dq.data_ptr(), dk.data_ptr(),
+                            dv.data_ptr(), ds.data_ptr(), bw, h, n, d, nw,
+                            ctx.scale, _dt(qc), _stream())
+        dbias = ds.sum(dim=0) if ctx.bias_requires else None
+        return dq, dk, dv, dbias, None, None
+
+
+def window_attention(q, k, v, bias, mask, scale, dropout=None):
+    """Swin window MHSA (K3): fused HIP kernels on GPU — eval forward AND
+    the training fwd+bwd (dropout-free, the Swin-ReID default); eager
+    composition on CPU or when attention dropout is active."""
+    drop_p = 0.0
+    if dropout is not None:
+        drop_p = float(getattr(dropout, "p", 0.0))
+    if (q.is_cuda and q.shape[-2] <= 64 and q.shape[-1] <= 64
+            and drop_p == 0.0 and extension_available()):
+        if not torch.is_grad_enabled():
+            ext = _ext_or_raise("window_attn_fwd")
+            out, *_ = _window_attn_fwd_raw(ext, q, k, v, bias, mask, scale)
             return out
+        return _WindowAttnFn.apply(q, k, v, bias, mask, scale)
     return ref.window_attention(q, k, v, bias, mask, scale, dropout)
+
+
+class _PatchMergeLNFn(torch.autograd.Function):
+    """Fused Swin PatchMerging gather + LayerNorm (K4, patch_merge.hip):
+    the 2×2 strided concat tensor never materialises; forward emits the
+    normalized [B, L/4, 4C] rows the reduction GEMM consumes, backward
+    scatters dx and reduces per-block dgamma/dbeta partials."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, H, W, eps):
+        ext = _ext_or_raise("patch_merge_ln_fwd")
+        x_c = x.detach().contiguous()
+        b, L, c = x_c.shape
+        rows = b * (H // 2) * (W // 2)
+        y = torch.empty(b, L // 4, 4 * c, device=x.device, dtype=x.dtype)
+        mean = torch.empty(rows, device=x.device, dtype=torch.float32)
+        rstd = torch.empty(rows, device=x.device, dtype=torch.float32)
+        gamma_c = gamma.detach().float().contiguous()
+        beta_c = beta.detach().float().contiguous()
+        ext.patch_merge_ln_fwd(x_c.data_ptr(), gamma_c.data_ptr(),
+                               beta_c.data_ptr(), y.data_ptr(),
+                               mean.data_ptr(), rstd.data_ptr(), rows, c, H,
+                               W, float(eps), _dt(x_c), _stream())
+        ctx.save_for_backward(x_c, gamma_c, mean, rstd)
+        ctx.hw = (H, W)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _ext_or_raise("patch_merge_ln_bwd")
+        x_c, gamma_c, mean, rstd = ctx.saved_tensors
+        H, W = ctx.hw
+        b, L, c = x_c.shape
+        rows = b * (H // 2) * (W // 2)
+        nblk = (rows + 3) // 4
+        dy_c = dy.contiguous().to(x_c.dtype)
+        dx = torch.empty_like(x_c)
+        dg_part = torch.empty(nblk, 4 * c, device=x_c.device,
+                              dtype=torch.float32)
+        db_part = torch.empty(nblk, 4 * c, device=x_c.device,
+                              dtype=torch.float32)
+        ext.patch_merge_ln_bwd(x_c.data_ptr(), gamma_c.data_ptr(),
+                               dy_c.data_ptr(), mean.data_ptr(),
+                               rstd.data_ptr(), dx.data_ptr(),
+                               dg_part.data_ptr(), db_part.data_ptr(), rows,
+                               c, H, W, _dt(x_c), _stream())
+        dgamma = dg_part.sum(0) if ctx.needs_input_grad[1] else None
+        dbeta = db_part.sum(0) if ctx.needs_input_grad[2] else None
+        return dx, dgamma, dbeta, None, None, None
+
+
+def patch_merge_ln(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+                   H: int, W: int, eps: float = 1e-5) -> Optional[torch.Tensor]:
+    """Fused PatchMerging gather+LN; None when out of regime (caller runs
+    the eager concat + LayerNorm)."""
+    if (not x.is_cuda or x.dim() != 3 or H % 2 or W % 2
+            or 4 * x.shape[2] > 3072 or x.shape[1] != H * W
+            or not extension_available()):
+        return None
+    if x.dtype not in (torch.float32, torch.bfloat16):
+        return None
+    return _PatchMergeLNFn.apply(x, gamma, beta, H, W, eps)
 
 
 def adaptive_linear_fwd(x: torch.Tensor, gw: torch.Tensor,
@@ -755,8 +864,8 @@ class _Conv3x3Fn(torch.autograd.Function):
         else:
             theta = _cl(weight.detach())
         k = weight.shape[0]
-        y = torch.empty(n, k, h, w, device=x.device, dtype=torch.bfloat16
-                        ).to(memory_format=torch.channels_last)
+        y = torch.empty(n, k, h, w, device=x.device, dtype=torch.bfloat16,
+                        memory_format=torch.channels_last)
         ext.conv3x3_img_fwd(x_bf.data_ptr(), theta.data_ptr(), y.data_ptr(),
                             n, h, w, c, k, _stream())
         ctx.save_for_backward(x_bf, theta)
@@ -773,19 +882,23 @@ class _Conv3x3Fn(torch.autograd.Function):
         dy_bf = _cl(dy.to(torch.bfloat16))
         dx = None
         if ctx.needs_input_grad[0]:
-            wt = torch.empty(c, k, 3, 3, device=dy.device, dtype=torch.bfloat16
-                             ).to(memory_format=torch.channels_last)
+            wt = torch.empty(c, k, 3, 3, device=dy.device,
+                             dtype=torch.bfloat16,
+                             memory_format=torch.channels_last)
             ext.conv3x3_wflip(theta.data_ptr(), wt.data_ptr(), c, k, _stream())
-            dx = torch.empty(n, c, h, w, device=dy.device, dtype=torch.bfloat16
-                             ).to(memory_format=torch.channels_last)
+            dx = torch.empty(n, c, h, w, device=dy.device,
+                             dtype=torch.bfloat16,
+                             memory_format=torch.channels_last)
             ext.conv3x3_img_fwd(dy_bf.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                                 n, h, w, k, c, _stream())
             if ctx.x_dtype != torch.bfloat16:
                 dx = dx.to(ctx.x_dtype)
         d_weight = d_aw = None
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[3]:
-            dw = torch.empty(k, c, 3, 3, device=dy.device, dtype=torch.float32
-                             ).to(memory_format=torch.channels_last)
+            # zeroed: the wgrad kernel's M-splits accumulate atomically
+            dw = torch.zeros(k, c, 3, 3, device=dy.device,
+                             dtype=torch.float32,
+                             memory_format=torch.channels_last)
             ext.conv3x3_wgrad(dy_bf.data_ptr(), x_bf.data_ptr(), dw.data_ptr(),
                               n, h, w, c, k, _stream())
             if ctx.needs_input_grad[3]:

@@ -82,9 +82,12 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
   }
 
   // ---- per-thread staging slots ----
+  // Loads are UNCONDITIONAL from clamped addresses; invalid slots select
+  // zero at the LDS store.  Branching around each load makes hipcc emit a
+  // vmcnt(0) wait per load — serial L2 round trips (guide §5 trap c).
   // x: slot = cell*4 + j  (cell = output pixel, j = 16B c-chunk)
   const int xs_slots = HW * 4;
-  int x_cell[2], x_j[2];
+  bool x_valid[2];
   int64_t x_gaddr[2];
   int x_laddr[2];
   {
@@ -93,22 +96,22 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
     for (int i = 0; i < 2; ++i) {
       const int slot = tid + 256 * i;
       const int cell = slot >> 2, j = slot & 3;
-      x_cell[i] = cell;
-      x_j[i] = j;
-      if (slot < xs_slots) {
+      x_valid[i] = slot < xs_slots;
+      if (x_valid[i]) {
         const int h = cell / Wd, w = cell % Wd;
         x_gaddr[i] = img_base + (int64_t)cell * C + j * 8;
         x_laddr[i] = ((h + 1) * Wp + (w + 1)) * CI_CS + ci_pbase(j);
       } else {
-        x_gaddr[i] = -1;
-        x_laddr[i] = 0;
+        x_gaddr[i] = img_base;   // clamped safe address
+        x_laddr[i] = ci_pbase(j);  // pad cell (0,0): zero writes keep it zero
       }
     }
   }
   // w: 9 slots, one per tap: k = tid>>2, j = tid&3
   const int wk = tid >> 2, wj = tid & 3;
   const bool w_ok = (k0 + wk) < K;
-  int64_t w_gbase = ((int64_t)(k0 + wk) * 9) * C + wj * 8;  // + tap*C + ct*32
+  const int64_t w_gbase =
+      ((int64_t)(w_ok ? (k0 + wk) : 0) * 9) * C + wj * 8;  // clamped
   const int w_lbase = wk * CI_CS + ci_pbase(wj);
 
   const int NT = C / CI_BK;
@@ -120,26 +123,23 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
     const int64_t coff = (int64_t)ct * CI_BK;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      if (x_gaddr[i] >= 0) {
-        xr[i] = *(const u16x8*)(X + x_gaddr[i] + coff);
-      }
+      xr[i] = *(const u16x8*)(X + x_gaddr[i] + coff);
     }
 #pragma unroll
     for (int t = 0; t < 9; ++t) {
-      if (w_ok) {
-        wr[t] = *(const u16x8*)(W + w_gbase + (int64_t)t * C + coff);
-      }
+      wr[t] = *(const u16x8*)(W + w_gbase + (int64_t)t * C + coff);
     }
   };
 
   auto store_tile = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      if (x_gaddr[i] >= 0) {
-        __hip_bfloat16* dst = &lx[buf][x_laddr[i]];
-        *(uint64_t*)dst = *(const uint64_t*)&xr[i];      // phys p..p+3
-        *(uint64_t*)(dst + 8) = *(((const uint64_t*)&xr[i]) + 1);  // p+8..p+11
-      }
+      __hip_bfloat16* dst = &lx[buf][x_laddr[i]];
+      const uint64_t lo = x_valid[i] ? *(const uint64_t*)&xr[i] : 0ull;
+      const uint64_t hi =
+          x_valid[i] ? *(((const uint64_t*)&xr[i]) + 1) : 0ull;
+      *(uint64_t*)dst = lo;          // phys p..p+3
+      *(uint64_t*)(dst + 8) = hi;    // p+8..p+11
     }
 #pragma unroll
     for (int t = 0; t < 9; ++t) {
@@ -281,122 +281,145 @@ extern "C" void flreid_conv3x3_wflip(const void* W, void* WT, int C, int K,
 
 constexpr int WG_BK = 64;    // k rows per block
 constexpr int WG_BC = 128;   // c cols per block
-constexpr int WG_BM = 32;    // m chunk
+constexpr int WG_BM = 64;    // m rows per iteration (two 32-m MFMA halves)
+constexpr int WG_MS = 72;    // padded m-stride (elems); 144 B, 16B-aligned
+constexpr int WG_SPLITM = 2; // M split across blocks (atomic accumulate)
 
-__device__ __forceinline__ int wg_perm(int m) {
-  return ((m & 15) >> 2) * 8 + (m >> 4) * 4 + (m & 3);
+__device__ __forceinline__ int wg_pm(int m) {
+  // per-32 fragment permutation, two halves side by side
+  return (m >> 5) * 32 + ((m & 15) >> 2) * 8 + ((m >> 4) & 1) * 4 + (m & 3);
 }
 
 __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
     const __hip_bfloat16* __restrict__ DY, const __hip_bfloat16* __restrict__ X,
     float* __restrict__ DW, int NB, int H, int Wd, int C, int K) {
-  __shared__ __hip_bfloat16 ldy[2][WG_BK * CI_CS];
-  __shared__ __hip_bfloat16 lxc[2][WG_BC * CI_CS];
+  __shared__ __hip_bfloat16 ldy[2][WG_BK * WG_MS];
+  __shared__ __hip_bfloat16 lxc[2][WG_BC * WG_MS];
 
   const int kb = blockIdx.x;          // k-block first: XCD dy-affinity
   const int cb = blockIdx.y;
-  const int tap = blockIdx.z;
+  const int tap = blockIdx.z % 9;
+  const int sp = blockIdx.z / 9;      // M-split index
   const int r = tap / 3 - 1, s = tap % 3 - 1;
   const int k0 = kb * WG_BK, c0 = cb * WG_BC;
   const int HW = H * Wd;
-  const int64_t M = (int64_t)NB * HW;
+  const int M = NB * HW;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wk = wave & 1, wc = wave >> 1;   // 2(k) × 2(c) wave grid
+  const int wk = wave & 1, wc = wave >> 1;   // 2(k) x 2(c) wave grid
   const int kg = lane >> 4;
 
-  // staging slots --------------------------------------------------------
-  // dy: 32m × 64k in 16B units = 256 slots: m_l = tid>>3, j = tid&7
-  const int dy_m = tid >> 3, dy_j = tid & 7;
-  // x: 32m × 128c = 512 slots: 2 per thread
-  const int x_m[2] = {tid >> 4, (tid + 256) >> 4};
-  const int x_j[2] = {tid & 15, (tid + 256) & 15};
+  // staging slots (loads unconditional from clamped addresses; invalid
+  // slots select zero at the LDS store — guide §5 trap c)
+  const int dy_m[2] = {tid >> 3, (tid + 256) >> 3};
+  const int dy_j[2] = {tid & 7, (tid + 256) & 7};
+  int x_m[4], x_j[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    x_m[i] = (tid + 256 * i) >> 4;
+    x_j[i] = (tid + 256 * i) & 15;
+  }
 
-  u16x8 dyr, xr[2];
-  bool dy_v = false, x_v[2] = {false, false};
+  u16x8 dyr[2], xr[4];
+  bool dy_v[2], x_v[4];
 
-  auto load_tile = [&](int64_t m0) {
-    {
-      const int64_t m = m0 + dy_m;
-      dy_v = m < M && (k0 + dy_j * 8) < K;
-      dyr = dy_v ? *(const u16x8*)(DY + m * K + k0 + dy_j * 8) : u16x8{};
-    }
+  auto load_tile = [&](int m0) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int64_t m = m0 + x_m[i];
-      bool ok = m < M;
+      const int m = m0 + dy_m[i];
+      dy_v[i] = m < M && (k0 + dy_j[i] * 8) < K;
+      const int64_t src = dy_v[i]
+          ? (int64_t)m * K + k0 + dy_j[i] * 8 : (int64_t)0;
+      dyr[i] = *(const u16x8*)(DY + src);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int m = m0 + x_m[i];
+      bool ok = m < M && (c0 + x_j[i] * 8) < C;
       int64_t src = 0;
       if (ok) {
-        const int n = (int)(m / HW);
-        const int rem = (int)(m % HW);
-        const int h = rem / Wd + r, w = rem % Wd + s;
+        const int n = m / HW;
+        const int rem = m - n * HW;
+        const int h = rem / Wd + r, w = rem - (rem / Wd) * Wd + s;
         ok = h >= 0 && h < H && w >= 0 && w < Wd;
-        src = ((int64_t)n * HW + h * Wd + w) * C + c0 + x_j[i] * 8;
-        ok = ok && (c0 + x_j[i] * 8) < C;
+        src = ok ? ((int64_t)n * HW + h * Wd + w) * C + c0 + x_j[i] * 8
+                 : (int64_t)0;
       }
       x_v[i] = ok;
-      xr[i] = ok ? *(const u16x8*)(X + src) : u16x8{};
+      xr[i] = *(const u16x8*)(X + src);
     }
   };
 
   auto store_tile = [&](int buf) {
-    {
-      // transposed: 8 k-rows, column p(m)
-      const int pm = wg_perm(dy_m);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int pm = wg_pm(dy_m[i]);
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        ldy[buf][(dy_j * 8 + e) * CI_CS + pm] =
-            dy_v ? *((const __hip_bfloat16*)&dyr + e) : __float2bfloat16(0.f);
+        ldy[buf][(dy_j[i] * 8 + e) * WG_MS + pm] =
+            dy_v[i] ? *((const __hip_bfloat16*)&dyr[i] + e)
+                    : __float2bfloat16(0.f);
       }
     }
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const int pm = wg_perm(x_m[i]);
+    for (int i = 0; i < 4; ++i) {
+      const int pm = wg_pm(x_m[i]);
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        lxc[buf][(x_j[i] * 8 + e) * CI_CS + pm] =
-            x_v[i] ? *((const __hip_bfloat16*)&xr[i] + e) : __float2bfloat16(0.f);
+        lxc[buf][(x_j[i] * 8 + e) * WG_MS + pm] =
+            x_v[i] ? *((const __hip_bfloat16*)&xr[i] + e)
+                   : __float2bfloat16(0.f);
       }
     }
   };
 
   if32x4 acc[2][4] = {};
 
-  const int64_t NIT = (M + WG_BM - 1) / WG_BM;
-  load_tile(0);
+  // this split's chunk range
+  const int NC = (M + WG_BM - 1) / WG_BM;
+  const int per = (NC + WG_SPLITM - 1) / WG_SPLITM;
+  const int ch0 = sp * per;
+  const int ch1 = min(NC, ch0 + per);
+  if (ch0 >= ch1) return;
+
+  load_tile(ch0 * WG_BM);
   store_tile(0);
-  if (NIT > 1) load_tile(WG_BM);
+  if (ch0 + 1 < ch1) load_tile((ch0 + 1) * WG_BM);
   __syncthreads();
 
-  for (int64_t it = 0; it < NIT; ++it) {
-    if (it + 1 < NIT) {
-      store_tile((int)((it + 1) & 1));
-      if (it + 2 < NIT) load_tile((it + 2) * WG_BM);
+  for (int it = ch0; it < ch1; ++it) {
+    const int buf = (it - ch0) & 1;
+    if (it + 1 < ch1) {
+      store_tile(buf ^ 1);
+      if (it + 2 < ch1) load_tile((it + 2) * WG_BM);
     }
-    const __hip_bfloat16* db = ldy[it & 1];
-    const __hip_bfloat16* xb = lxc[it & 1];
-    ibf16x8 afrag[2];
+    const __hip_bfloat16* db = ldy[buf];
+    const __hip_bfloat16* xb = lxc[buf];
 #pragma unroll
-    for (int fk = 0; fk < 2; ++fk) {
-      afrag[fk] = *(const ibf16x8*)(
-          db + (wk * 32 + fk * 16 + (lane & 15)) * CI_CS + 8 * kg);
-    }
-#pragma unroll
-    for (int fc = 0; fc < 4; ++fc) {
-      const ibf16x8 bfrag = *(const ibf16x8*)(
-          xb + (wc * 64 + fc * 16 + (lane & 15)) * CI_CS + 8 * kg);
+    for (int hh = 0; hh < 2; ++hh) {       // two 32-m halves per tile
+      ibf16x8 afrag[2];
 #pragma unroll
       for (int fk = 0; fk < 2; ++fk) {
-        acc[fk][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[fk], bfrag, acc[fk][fc], 0, 0, 0);
+        afrag[fk] = *(const ibf16x8*)(
+            db + (wk * 32 + fk * 16 + (lane & 15)) * WG_MS + hh * 32 + 8 * kg);
+      }
+#pragma unroll
+      for (int fc = 0; fc < 4; ++fc) {
+        const ibf16x8 bfrag = *(const ibf16x8*)(
+            xb + (wc * 64 + fc * 16 + (lane & 15)) * WG_MS + hh * 32 + 8 * kg);
+#pragma unroll
+        for (int fk = 0; fk < 2; ++fk) {
+          acc[fk][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[fk], bfrag, acc[fk][fc], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
   }
 
-  // D rows = k (A), cols = c (B)
+  // D rows = k (A), cols = c (B); accumulate across the M splits
 #pragma unroll
   for (int fk = 0; fk < 2; ++fk) {
 #pragma unroll
@@ -406,7 +429,7 @@ __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
         const int k = k0 + wk * 32 + fk * 16 + (lane >> 4) * 4 + reg;
         const int c = c0 + wc * 64 + fc * 16 + (lane & 15);
         if (k < K && c < C) {
-          DW[((int64_t)k * 9 + tap) * C + c] = acc[fk][fc][reg];
+          atomicAdd(&DW[((int64_t)k * 9 + tap) * C + c], acc[fk][fc][reg]);
         }
       }
     }
@@ -419,7 +442,8 @@ extern "C" void flreid_conv3x3_wgrad(const void* DY, const void* X, float* DW,
   if (C % 8 || K % 8) {
     throw std::runtime_error("conv3x3_wgrad: C%8 or K%8 != 0");
   }
-  dim3 grid((K + WG_BK - 1) / WG_BK, (C + WG_BC - 1) / WG_BC, 9);
+  // DW must be ZEROED by the caller (splits accumulate atomically)
+  dim3 grid((K + WG_BK - 1) / WG_BK, (C + WG_BC - 1) / WG_BC, 9 * WG_SPLITM);
   hipLaunchKernelGGL(conv3x3_wgrad_kernel, grid, dim3(256), 0, stream,
                      (const __hip_bfloat16*)DY, (const __hip_bfloat16*)X, DW,
                      NB, H, Wd, C, K);

@@ -29,6 +29,11 @@ extern "C" void flreid_window_attn_fwd(const void*, const void*, const void*,
                                        const float*, const float*, void*,
                                        int64_t, int, int, int, int, float,
                                        int, hipStream_t);
+extern "C" void flreid_window_attn_bwd(const void*, const void*, const void*,
+                                       const float*, const float*,
+                                       const void*, void*, void*, void*,
+                                       float*, int64_t, int, int, int, int,
+                                       float, int, hipStream_t);
 extern "C" void flreid_triplet_fwd(const float*, const float*, const int64_t*,
                                    float*, int*, int*, int, int, float,
                                    hipStream_t);
@@ -64,6 +69,15 @@ extern "C" void flreid_conv3x3_wflip(const void*, void*, int, int,
                                      hipStream_t);
 extern "C" void flreid_conv3x3_wgrad(const void*, const void*, float*, int,
                                      int, int, int, int, hipStream_t);
+extern "C" void flreid_patch_merge_ln_fwd(const void*, const float*,
+                                          const float*, void*, float*,
+                                          float*, int64_t, int, int, int,
+                                          float, int, hipStream_t);
+extern "C" void flreid_patch_merge_ln_bwd(const void*, const float*,
+                                          const void*, const float*,
+                                          const float*, void*, float*,
+                                          float*, int64_t, int, int, int,
+                                          int, hipStream_t);
 extern "C" void flreid_kd_fwd(const float*, const float*, float*, float*,
                               int64_t, int64_t, float, hipStream_t);
 extern "C" void flreid_icarl_distill(const float*, const int64_t*,
@@ -138,6 +152,18 @@ PYBIND11_MODULE(_flreid_hip, m) {
               (const void*)q, (const void*)k, (const void*)v,
               (const float*)bias, (const float*)mask, (void*)out, BW, H, N, D,
               nW, scale, dtype, as_stream(stream));
+        });
+
+  m.def("window_attn_bwd",
+        [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t bias,
+           uintptr_t mask, uintptr_t dout, uintptr_t dq, uintptr_t dk,
+           uintptr_t dv, uintptr_t ds, int64_t BW, int H, int N, int D,
+           int nW, float scale, int dtype, uintptr_t stream) {
+          flreid::flreid_window_attn_bwd(
+              (const void*)q, (const void*)k, (const void*)v,
+              (const float*)bias, (const float*)mask, (const void*)dout,
+              (void*)dq, (void*)dk, (void*)dv, (float*)ds, BW, H, N, D, nW,
+              scale, dtype, as_stream(stream));
         });
 
   m.def("triplet_fwd",
@@ -243,6 +269,26 @@ PYBIND11_MODULE(_flreid_hip, m) {
           flreid::flreid_conv3x3_wgrad((const void*)dy, (const void*)x,
                                        (float*)dw, NB, H, Wd, C, K,
                                        as_stream(stream));
+        });
+
+  m.def("patch_merge_ln_fwd",
+        [](uintptr_t x, uintptr_t gamma, uintptr_t beta, uintptr_t y,
+           uintptr_t mean, uintptr_t rstd, int64_t rows, int C, int H,
+           int W, float eps, int dtype, uintptr_t stream) {
+          flreid::flreid_patch_merge_ln_fwd(
+              (const void*)x, (const float*)gamma, (const float*)beta,
+              (void*)y, (float*)mean, (float*)rstd, rows, C, H, W, eps,
+              dtype, as_stream(stream));
+        });
+
+  m.def("patch_merge_ln_bwd",
+        [](uintptr_t x, uintptr_t gamma, uintptr_t dy, uintptr_t mean,
+           uintptr_t rstd, uintptr_t dx, uintptr_t dg, uintptr_t db,
+           int64_t rows, int C, int H, int W, int dtype, uintptr_t stream) {
+          flreid::flreid_patch_merge_ln_bwd(
+              (const void*)x, (const float*)gamma, (const void*)dy,
+              (const float*)mean, (const float*)rstd, (void*)dx, (float*)dg,
+              (float*)db, rows, C, H, W, dtype, as_stream(stream));
         });
 
   m.def("kd_fwd",

@@ -608,3 +608,94 @@ def test_compose2_bf16_channels_last():
     assert out.dtype == torch.bfloat16
     expected = (atten.view(1, 1, 1, 3) * gw + aw).bfloat16()
     assert torch.allclose(out.float(), expected.float(), atol=2e-2)
+
+
+def test_window_attention_training_bwd_matches_eager():
+    """K3 TRAINING path: fused fwd+bwd (window_attn.hip bwd kernel) vs the
+    eager fp32 composition — dQ/dK/dV and the bias gradient."""
+    torch.manual_seed(4)
+    bw, h, n, d, nw = 8, 3, 49, 32, 4
+    q = torch.randn(bw, h, n, d, device="cuda", requires_grad=True)
+    k = torch.randn(bw, h, n, d, device="cuda", requires_grad=True)
+    v = torch.randn(bw, h, n, d, device="cuda", requires_grad=True)
+    bias = torch.randn(h, n, n, device="cuda", requires_grad=True)
+    mask = (torch.randn(nw, n, n, device="cuda") > 0).float() * -100.0
+    scale = d ** -0.5
+
+    out = ops.window_attention(q, k, v, bias, mask, scale)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    grads = [q.grad.clone(), k.grad.clone(), v.grad.clone(), bias.grad.clone()]
+    for t in (q, k, v, bias):
+        t.grad = None
+    ref_out = ref.window_attention(q, k, v, bias, mask, scale, None)
+    ref_out.backward(g)
+
+    assert torch.allclose(out, ref_out, atol=1e-4, rtol=1e-4)
+    for got, t, name in zip(grads, (q, k, v, bias), "qkvb"):
+        assert torch.allclose(got, t.grad, atol=1e-3, rtol=1e-3), \
+            (name, (got - t.grad).abs().max())
+
+
+def test_window_attention_training_bwd_bf16():
+    """bf16 I/O training path sanity (autocast regime)."""
+    torch.manual_seed(5)
+    bw, h, n, d = 4, 3, 49, 32
+    q = torch.randn(bw, h, n, d, device="cuda").bfloat16().requires_grad_(True)
+    k = torch.randn(bw, h, n, d, device="cuda").bfloat16().requires_grad_(True)
+    v = torch.randn(bw, h, n, d, device="cuda").bfloat16().requires_grad_(True)
+    bias = torch.randn(h, n, n, device="cuda", requires_grad=True)
+    out = ops.window_attention(q, k, v, bias, None, d ** -0.5)
+    out.sum().backward()
+
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    bf = bias.detach().clone().requires_grad_(True)
+    ref.window_attention(qf, kf, vf, bf, None, d ** -0.5, None).sum().backward()
+    assert torch.allclose(q.grad.float(), qf.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(bias.grad, bf.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_patch_merge_ln_fwd_bwd_matches_eager():
+    """K4 fused gather+LN (patch_merge.hip) vs the eager concat+LayerNorm
+    composition in fp32 — output, dx, dgamma, dbeta."""
+    torch.manual_seed(6)
+    b, h, w, c = 3, 8, 4, 96
+    x = torch.randn(b, h * w, c, device="cuda", requires_grad=True)
+    gamma = torch.randn(4 * c, device="cuda", requires_grad=True)
+    beta = torch.randn(4 * c, device="cuda", requires_grad=True)
+    out = ops.patch_merge_ln(x, gamma, beta, h, w)
+    assert out is not None and out.shape == (b, h * w // 4, 4 * c)
+    g = torch.randn_like(out)
+    out.backward(g)
+    got = [x.grad.clone(), gamma.grad.clone(), beta.grad.clone()]
+    for t in (x, gamma, beta):
+        t.grad = None
+
+    xv = x.view(b, h, w, c)
+    cat = torch.cat([xv[:, 0::2, 0::2], xv[:, 1::2, 0::2],
+                     xv[:, 0::2, 1::2], xv[:, 1::2, 1::2]], dim=-1)
+    ref_out = torch.nn.functional.layer_norm(
+        cat.reshape(b, -1, 4 * c), (4 * c,), gamma, beta)
+    ref_out.backward(g)
+    assert torch.allclose(out, ref_out, atol=1e-4, rtol=1e-4)
+    for got_g, t, name in zip(got, (x, gamma, beta), ("dx", "dg", "db")):
+        assert torch.allclose(got_g, t.grad, atol=1e-3, rtol=1e-3), \
+            (name, (got_g - t.grad).abs().max())
+
+
+def test_patch_merge_ln_bf16():
+    torch.manual_seed(7)
+    b, h, w, c = 2, 4, 4, 32
+    x = torch.randn(b, h * w, c, device="cuda").bfloat16()
+    gamma = torch.ones(4 * c, device="cuda")
+    beta = torch.zeros(4 * c, device="cuda")
+    out = ops.patch_merge_ln(x, gamma, beta, h, w)
+    xv = x.float().view(b, h, w, c)
+    cat = torch.cat([xv[:, 0::2, 0::2], xv[:, 1::2, 0::2],
+                     xv[:, 0::2, 1::2], xv[:, 1::2, 1::2]], dim=-1)
+    expected = torch.nn.functional.layer_norm(
+        cat.reshape(b, -1, 4 * c), (4 * c,), gamma, beta)
+    assert torch.allclose(out.float(), expected, atol=5e-2, rtol=5e-2)
