@@ -144,29 +144,21 @@ class AdaptiveConv2d(AdaptiveBase):
                 and gw.shape[3] == 1 and _is_one(self.stride)
                 and _is_zero(self.padding)):
             # pointwise conv == GEMM over N·H·W; on channels-last input the
-            # [NHW, C] view is free.  Two routes (bwd always hipBLASLt —
-            # MIOpen picked a naive weight-grad kernel for these shapes):
-            #  - OPT-IN (FLREID_FUSED_1X1=1) fused K2 MFMA GEMM with
-            #    compose-in-prologue (θ never hits HBM): the standalone probe
-            #    (benchmarks/adaptive_1x1_probe.py) shows wins at K≤512 /
-            #    K≤1024·N≤1024, but the end-to-end effect (~+1%) sits below
-            #    the box-to-box noise floor, so the proven path stays default;
-            #  - compose kernel + hipBLASLt otherwise.
+            # [NHW, C] view is free.  DEFAULT route (K2): θ composed
+            # straight to bf16 (compose2 — one pass, no fp32 θ, no autocast
+            # cast) + bf16 hipBLASLt GEMMs fwd/bwd, aw's grad cast once to
+            # fp32.  Disable with FLREID_NO_FUSED_1X1=1.
             b, c, h, w = data.shape
             n_out = gw.shape[0]
             xv = data.permute(0, 2, 3, 1).reshape(-1, c)
-            if (not self.global_weight_atten.requires_grad
-                    and not gw.requires_grad and c % 32 == 0
-                    and (c <= 512 or (c <= 1024 and n_out <= 1024))
-                    and os.environ.get("FLREID_FUSED_1X1", "0") == "1"
-                    and ops.extension_available()):
-                # the K2 kernel's contract is atten[K]: broadcast the conv's
-                # scalar atten (last dim of the 4-D weight is 1) to length K
-                atten_k = self.global_weight_atten.expand(c).contiguous()
-                y = ops.adaptive_linear(xv, gw.view(n_out, c), atten_k,
-                                        self.adaptive_weight.view(n_out, c),
-                                        self.adaptive_bias)
-            else:
+            y = None
+            if not gw.requires_grad and not self.global_weight_atten.requires_grad:
+                aw2d = self.adaptive_weight.view(n_out, c) \
+                    if self.adaptive_weight.numel() else None
+                y = ops.adaptive_linear_1x1(xv, gw.view(n_out, c),
+                                            self.global_weight_atten, aw2d,
+                                            self.adaptive_bias)
+            if y is None:
                 theta = self.composed_weight().view(n_out, c)
                 y = F.linear(xv, theta, self.adaptive_bias)
             return y.view(b, h, w, -1).permute(0, 3, 1, 2)

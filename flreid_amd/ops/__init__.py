@@ -577,7 +577,12 @@ class _AdaptiveLinearFn(torch.autograd.Function):
     def backward(ctx, grad_out):
         x_bf, gw, atten, aw = ctx.saved_tensors
         g = grad_out.to(torch.bfloat16)
-        theta = adaptive_compose(gw, atten, aw).to(torch.bfloat16)
+        ext = _load_extension()
+        if ext is not None and gw.dtype == torch.float32:
+            theta = compose_theta_bf16(ext, gw.contiguous(), atten,
+                                       aw.contiguous())
+        else:
+            theta = adaptive_compose(gw, atten, aw).to(torch.bfloat16)
         grad_x = (g @ theta).to(ctx.x_dtype)
         grad_aw = (g.t() @ x_bf).to(aw.dtype)
         grad_bias = grad_out.sum(0).to(torch.float32) if ctx.has_bias else None
@@ -830,7 +835,10 @@ def compose_theta_bf16(ext, gw: torch.Tensor, atten: Optional[torch.Tensor],
         inner, L = gw_d.shape[1], gw_d.shape[-1]
     else:
         assert gw_d.is_contiguous()
-        inner, L = 1, gw_d.shape[-1]
+        inner = 1
+        L = atten.numel() if atten is not None else 1
+    if atten is not None:
+        assert atten.numel() == L
     out = torch.empty_like(gw_d, dtype=torch.bfloat16)
     ext.compose2(gw_d.data_ptr(),
                  atten.detach().float().contiguous().data_ptr() if atten is not None else 0,
@@ -838,6 +846,53 @@ def compose_theta_bf16(ext, gw: torch.Tensor, atten: Optional[torch.Tensor],
                  out.data_ptr(), gw_d.numel(), L, inner, _dt(gw_d), _BF16,
                  _stream())
     return out
+
+
+class _AdaptiveLinear1x1Fn(torch.autograd.Function):
+    """Default-on fused adaptive 1×1/linear path (K2 at M=N·H·W): θ is
+    composed straight to bf16 (compose2 — no fp32 θ in HBM, no autocast
+    cast pass) and the GEMMs run bf16 hipBLASLt.  aw's gradient is the
+    bf16 wgrad GEMM cast once to fp32.  gw/atten frozen (FedSTIL)."""
+
+    @staticmethod
+    def forward(ctx, x, gw2d, atten, aw2d, bias):
+        ext = _ext_or_raise("compose2")
+        x_bf = x.detach().to(torch.bfloat16).contiguous()
+        theta = compose_theta_bf16(ext, gw2d, atten, aw2d)
+        y = x_bf @ theta.t()
+        if bias is not None:
+            y = y + bias.detach().to(torch.bfloat16)
+        ctx.save_for_backward(x_bf, theta)
+        ctx.x_dtype = x.dtype
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x_bf, theta = ctx.saved_tensors
+        g = grad_out.to(torch.bfloat16)
+        dx = (g @ theta).to(ctx.x_dtype)
+        d_aw = (g.t() @ x_bf).to(torch.float32)
+        d_bias = grad_out.sum(0).to(torch.float32) if ctx.has_bias else None
+        return dx, None, None, d_aw, d_bias
+
+
+def adaptive_linear_1x1(x2d: torch.Tensor, gw2d: torch.Tensor,
+                        atten: Optional[torch.Tensor],
+                        aw2d: Optional[torch.Tensor],
+                        bias: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+    """Dispatcher for the fused 1×1 route; None when out of regime."""
+    if (not x2d.is_cuda or not extension_available()
+            or os.environ.get("FLREID_NO_FUSED_1X1", "0") == "1"):
+        return None
+    if gw2d.requires_grad or (atten is not None and atten.requires_grad):
+        return None
+    if x2d.dtype != torch.bfloat16 and not torch.is_autocast_enabled():
+        return None
+    if gw2d.dtype != torch.float32 or (aw2d is not None
+                                       and aw2d.dtype != torch.float32):
+        return None
+    return _AdaptiveLinear1x1Fn.apply(x2d, gw2d, atten, aw2d, bias)
 
 
 class _Conv3x3Fn(torch.autograd.Function):

@@ -450,8 +450,8 @@ def test_l1_drift_fused_matches_foreach():
 
 
 def test_adaptive_conv1x1_fused_route(monkeypatch):
-    """Opt-in FLREID_FUSED_1X1=1 pointwise route (K2 GEMM with scalar atten,
-    compose-in-prologue) vs the compose+hipBLASLt default."""
+    """Default-on fused pointwise route (compose2-to-bf16 + bf16 hipBLASLt,
+    _AdaptiveLinear1x1Fn) vs the eager compose+linear fallback."""
     import copy
     from flreid_amd.models.adaptive import AdaptiveConv2d
     torch.manual_seed(4)
@@ -464,12 +464,11 @@ def test_adaptive_conv1x1_fused_route(monkeypatch):
     x2 = x.detach().clone().requires_grad_(True)
 
     # both routes run under bf16 autocast, like the training round
-    monkeypatch.setenv("FLREID_FUSED_1X1", "1")
     with torch.autocast("cuda", dtype=torch.bfloat16):
-        y = conv(x)
-    monkeypatch.setenv("FLREID_FUSED_1X1", "0")
+        y = conv(x)                       # default: fused route
+    monkeypatch.setenv("FLREID_NO_FUSED_1X1", "1")
     with torch.autocast("cuda", dtype=torch.bfloat16):
-        y2 = conv2(x2)
+        y2 = conv2(x2)                    # eager compose + F.linear
     assert torch.allclose(y.float(), y2.float(), atol=0.1, rtol=5e-2)
 
     dy = torch.randn_like(y2.float())
