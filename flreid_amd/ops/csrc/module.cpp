@@ -67,6 +67,8 @@ extern "C" void flreid_conv3x3_img_fwd(const void*, const void*, void*, int,
                                        int, int, int, int, hipStream_t);
 extern "C" void flreid_conv3x3_wflip(const void*, void*, int, int,
                                      hipStream_t);
+extern "C" void flreid_conv3x3_tile(const void*, const float*, const void*,
+                                    void*, int, int, int, int, hipStream_t);
 extern "C" void flreid_conv3x3_wgrad(const void*, const void*, float*, int,
                                      int, int, int, int, hipStream_t);
 extern "C" void flreid_patch_merge_ln_fwd(const void*, const float*,
@@ -261,6 +263,14 @@ PYBIND11_MODULE(_flreid_hip, m) {
         [](uintptr_t w, uintptr_t wt, int C, int K, uintptr_t stream) {
           flreid::flreid_conv3x3_wflip((const void*)w, (void*)wt, C, K,
                                        as_stream(stream));
+        });
+
+  m.def("conv3x3_tile",
+        [](uintptr_t gw, uintptr_t atten, uintptr_t aw, uintptr_t out,
+           int C, int K, int in_dtype, int mode, uintptr_t stream) {
+          flreid::flreid_conv3x3_tile((const void*)gw, (const float*)atten,
+                                      (const void*)aw, (void*)out, C, K,
+                                      in_dtype, mode, as_stream(stream));
         });
 
   m.def("conv3x3_wgrad",
