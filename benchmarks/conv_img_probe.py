@@ -64,11 +64,18 @@ def run_shape(n, c, h, w, k):
         return dx
 
     def ours_wgrad():
-        dw = torch.empty(k, c, 3, 3, device="cuda", dtype=torch.float32,
-                         memory_format=torch.channels_last).zero_()
-        ext.conv3x3_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr(),
+        part = torch.empty(2, k * 9 * c, device="cuda", dtype=torch.float32)
+        ext.conv3x3_wgrad(dy.data_ptr(), x.data_ptr(), part.data_ptr(),
                           n, h, w, c, k, torch.cuda.current_stream().cuda_stream)
-        return dw
+        return (part[0] + part[1]).view(k, 3, 3, c).permute(0, 3, 1, 2)
+
+    def ours_fwd_ldsw():
+        y = torch.empty(n, k, h, w, device="cuda", dtype=torch.bfloat16,
+                        memory_format=torch.channels_last)
+        ext.conv3x3_img_fwd_ldsw(x.data_ptr(), wt_bf.data_ptr(), y.data_ptr(),
+                                 n, h, w, c, k,
+                                 torch.cuda.current_stream().cuda_stream)
+        return y
 
     def lib_dgrad():
         return torch.nn.grad.conv2d_input((n, c, h, w), wt_bf, dy, padding=1)
@@ -78,7 +85,8 @@ def run_shape(n, c, h, w, k):
 
     print(f"== {n}x{c}x{h}x{w} -> {k}  ({flops/1e9:.1f} GFLOP) ==")
     for name, fn in (("ours_tile", ours_tile),
-                     ("ours_fwd", ours_fwd), ("lib_fwd", lib_fwd),
+                     ("ours_fwd", ours_fwd),
+                     ("ours_fwd_ldsw", ours_fwd_ldsw), ("lib_fwd", lib_fwd),
                      ("ours_dgrad", ours_dgrad), ("lib_dgrad", lib_dgrad),
                      ("ours_wgrad", ours_wgrad), ("lib_wgrad", lib_wgrad)):
         try:

@@ -426,12 +426,15 @@ class Operator(BaseReIDOperator):
                             full=d.shape[0] == chunk)
                     # keep the autocast dtype (fp32 storage would only double
                     # HBM traffic — the tap was COMPUTED in bf16) and store
-                    # rows physically [H, W, C]: the tap is channels-last, so
-                    # the permute is the identity on memory and every
-                    # rehearsal batch becomes a free channels-last view.
-                    # clone() is REQUIRED: eval_graphed returns the graph's
-                    # static output buffer, overwritten by the next replay
-                    taps.append(tap.permute(0, 2, 3, 1).clone())
+                    # 4-D taps physically [H, W, C]: the tap is channels-last,
+                    # so the permute is the identity on memory and every
+                    # rehearsal batch becomes a free channels-last view
+                    # (Swin taps are 3-D token grids [B, L, C] — stored
+                    # as-is).  clone() is REQUIRED: eval_graphed returns the
+                    # graph's static output buffer, overwritten by the next
+                    # replay
+                    taps.append((tap.permute(0, 2, 3, 1) if tap.dim() == 4
+                                 else tap).clone())
                 del data_all, datas
             else:
                 taps, pids, classes = [], [], []

@@ -963,12 +963,14 @@ class _Conv3x3Fn(torch.autograd.Function):
                 dx = dx.to(ctx.x_dtype)
         d_weight = d_aw = None
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[3]:
-            # zeroed: the wgrad kernel's M-splits accumulate atomically
-            dw = torch.empty(k, c, 3, 3, device=dy.device,
-                             dtype=torch.float32,
-                             memory_format=torch.channels_last).zero_()
-            ext.conv3x3_wgrad(dy_bf.data_ptr(), x_bf.data_ptr(), dw.data_ptr(),
-                              n, h, w, c, k, _stream())
+            # two exclusive M-split partials, summed here (plain stores beat
+            # the measured ~0.4 ms of 4.7M fp32 atomicAdds)
+            part = torch.empty(2, k * 9 * c, device=dy.device,
+                               dtype=torch.float32)
+            ext.conv3x3_wgrad(dy_bf.data_ptr(), x_bf.data_ptr(),
+                              part.data_ptr(), n, h, w, c, k, _stream())
+            # [K][9][C] flat == the channels-last [K,C,3,3] physical layout
+            dw = (part[0] + part[1]).view(k, 3, 3, c).permute(0, 3, 1, 2)
             if ctx.needs_input_grad[3]:
                 d_aw = dw              # identity composition, fp32 direct
             else:

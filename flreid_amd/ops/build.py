@@ -15,7 +15,7 @@ import sysconfig
 CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)), "csrc")
 OUT = os.path.join(os.path.dirname(os.path.abspath(__file__)), "_flreid_hip.so")
 
-SOURCES = ["module.cpp", "elementwise.hip", "distance.hip", "window_attn.hip", "triplet.hip", "adaptive_gemm.hip", "conv3x3.hip", "conv3x3_img.hip", "bn_train.hip", "drift.hip", "kd.hip", "patch_merge.hip"]
+SOURCES = ["module.cpp", "elementwise.hip", "distance.hip", "window_attn.hip", "triplet.hip", "adaptive_gemm.hip", "conv3x3.hip", "conv3x3_img.hip", "conv3x3_img_ldsw.hip", "bn_train.hip", "drift.hip", "kd.hip", "patch_merge.hip"]
 
 
 def _pybind11_includes():

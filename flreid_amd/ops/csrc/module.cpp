@@ -69,6 +69,9 @@ extern "C" void flreid_conv3x3_wflip(const void*, void*, int, int,
                                      hipStream_t);
 extern "C" void flreid_conv3x3_tile(const void*, const float*, const void*,
                                     void*, int, int, int, int, hipStream_t);
+extern "C" void flreid_conv3x3_img_fwd_ldsw(const void*, const void*, void*,
+                                            int, int, int, int, int,
+                                            hipStream_t);
 extern "C" void flreid_conv3x3_wgrad(const void*, const void*, float*, int,
                                      int, int, int, int, hipStream_t);
 extern "C" void flreid_patch_merge_ln_fwd(const void*, const float*,
@@ -263,6 +266,14 @@ PYBIND11_MODULE(_flreid_hip, m) {
         [](uintptr_t w, uintptr_t wt, int C, int K, uintptr_t stream) {
           flreid::flreid_conv3x3_wflip((const void*)w, (void*)wt, C, K,
                                        as_stream(stream));
+        });
+
+  m.def("conv3x3_img_fwd_ldsw",
+        [](uintptr_t x, uintptr_t w, uintptr_t y, int NB, int H, int Wd,
+           int C, int K, uintptr_t stream) {
+          flreid::flreid_conv3x3_img_fwd_ldsw((const void*)x, (const void*)w,
+                                              (void*)y, NB, H, Wd, C, K,
+                                              as_stream(stream));
         });
 
   m.def("conv3x3_tile",
