@@ -919,8 +919,8 @@ class _Conv3x3Fn(torch.autograd.Function):
 
     fwd:   θ_bf16 = compose2(gw, atten, aw) (or a cast of a plain weight);
            y = conv3x3_img_fwd(x, θ) — one-image-per-block halo kernel.
-    dgrad: dx = conv3x3_img_fwd(dy, wflip(θ)) — same kernel, flipped
-           transposed taps.
+    dgrad: dx = conv3x3_img_fwd(dy, tile(θ, flipped)) — the same kernel on
+           the flip-transposed weight tile.
     wgrad: dθ = conv3x3_wgrad(dy, x), fp32 accumulate/output; for the
            adaptive layer dθ IS d(adaptive_weight) (identity composition),
            so no bf16 round-trip on the weight gradient.
@@ -963,11 +963,21 @@ class _Conv3x3Fn(torch.autograd.Function):
                 dx = dx.to(ctx.x_dtype)
         d_weight = d_aw = None
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[3]:
-            # two exclusive M-split partials, summed here (plain stores beat
-            # the measured ~0.4 ms of 4.7M fp32 atomicAdds)
+            # wgrad on pre-transposed operands (the in-kernel transposed
+            # scatter measured 75 LDS-conflict cycles/MFMA); two exclusive
+            # M-split partials summed here
+            m_rows = n * h * w
+            dyt = torch.empty(k, m_rows, device=dy.device,
+                              dtype=torch.bfloat16)
+            xt = torch.empty(c, m_rows, device=dy.device,
+                             dtype=torch.bfloat16)
+            ext.transpose_bf16(dy_bf.data_ptr(), dyt.data_ptr(), m_rows, k,
+                               _stream())
+            ext.transpose_bf16(x_bf.data_ptr(), xt.data_ptr(), m_rows, c,
+                               _stream())
             part = torch.empty(2, k * 9 * c, device=dy.device,
                                dtype=torch.float32)
-            ext.conv3x3_wgrad(dy_bf.data_ptr(), x_bf.data_ptr(),
+            ext.conv3x3_wgrad(dyt.data_ptr(), xt.data_ptr(),
                               part.data_ptr(), n, h, w, c, k, _stream())
             # [K][9][C] flat == the channels-last [K,C,3,3] physical layout
             dw = (part[0] + part[1]).view(k, 3, 3, c).permute(0, 3, 1, 2)

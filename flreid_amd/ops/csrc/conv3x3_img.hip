@@ -59,11 +59,12 @@ __device__ __forceinline__ int ci_pbase(int j) {
 
 // -------------------------------------------------------------------------
 // Weight tiling: θ is emitted by the composition kernel in a CONV-TILED
-// bf16 layout [C/32][9][K][32p] (p = the MFMA fragment permutation), so in
-// the conv kernel one global_load_dwordx4 per lane IS the B-fragment —
-// weights never touch LDS, and a wave's 64 lanes read 1 KB contiguous per
-// (c-tile, tap).  mode 1 produces the dgrad tile [K/32][9][C][32p] with
-// flipped taps (wT[c][2-r][2-s][k]).
+// bf16 layout [C/32][9][4kg][K][8] — per (c-tile, tap) panel the fragment
+// chunks are ordered (kg, k): lane (fn, kg) reads its 16-B B-fragment at
+// panel + (kg·K + fn)·8, so a wave's 64 lanes cover two contiguous 512-B
+// runs per quarter-wave group and weights never touch LDS.  mode 1
+// produces the dgrad tile with C/K swapped and flipped taps
+// (wT[c][2-r][2-s][k]).
 // -------------------------------------------------------------------------
 
 __device__ __forceinline__ int ci_invp(int p) {
@@ -79,18 +80,20 @@ __global__ __launch_bounds__(256) void conv3x3_tile_kernel(
   const int64_t total = (int64_t)C * 9 * K;
   for (int64_t o = (int64_t)blockIdx.x * 256 + threadIdx.x; o < total;
        o += (int64_t)gridDim.x * 256) {
-    const int p = (int)(o & 31);
-    const int64_t o2 = o >> 5;
+    const int e = (int)(o & 7);           // element within the 16-B chunk
+    const int64_t o2 = o >> 3;
     int k, c, tap;
-    if (mode == 0) {                      // [C/32][9][K][32p(C)]
+    if (mode == 0) {                      // [C/32][9][4kg][K][8]
       k = (int)(o2 % K);
-      tap = (int)((o2 / K) % 9);
-      c = (int)(o2 / ((int64_t)K * 9)) * 32 + ci_invp(p);
-    } else {                              // [K/32][9][C][32p(K)], flipped
+      const int kg = (int)((o2 / K) & 3);
+      tap = (int)((o2 / ((int64_t)K * 4)) % 9);
+      c = (int)(o2 / ((int64_t)K * 4 * 9)) * 32 + ci_invp(kg * 8 + e);
+    } else {                              // [K/32][9][4kg][C][8], flipped
       c = (int)(o2 % C);
-      const int ft = (int)((o2 / C) % 9);
+      const int kg = (int)((o2 / C) & 3);
+      const int ft = (int)((o2 / ((int64_t)C * 4)) % 9);
       tap = (2 - ft / 3) * 3 + (2 - ft % 3);
-      k = (int)(o2 / ((int64_t)C * 9)) * 32 + ci_invp(p);
+      k = (int)(o2 / ((int64_t)C * 4 * 9)) * 32 + ci_invp(kg * 8 + e);
     }
     const int64_t src = ((int64_t)k * 9 + tap) * C + c;
     const float a = ATTEN ? ATTEN[tap % 3] : 1.0f;
@@ -169,12 +172,13 @@ __global__ __launch_bounds__(256) void conv3x3_img_fwd_kernel(
   }
 
   // per-lane weight-fragment base in the tiled layout (elements):
-  // WT[((ct*9 + tap)*K + k0+fn)*32 + kg*8]
-  const int64_t w_lane = ((int64_t)(k0 + fn)) * 32 + kg * 8;
+  // WT[(((ct*9 + tap)*4 + kg)*K + k0+fn)*8] — monotonic in lane, two
+  // contiguous 512-B runs per quarter-wave
+  const int64_t w_lane = ((int64_t)kg * K + (k0 + fn)) * 8;
   const int64_t w_ctstride = (int64_t)9 * K * 32;
   const int64_t w_tapstride = (int64_t)K * 32;
   const bool w_ok = (k0 + fn) < K;
-  const int64_t w_base = w_ok ? w_lane : (int64_t)(kg * 8);
+  const int64_t w_base = w_ok ? w_lane : (int64_t)kg * K * 8;
 
   const int NT = C / CI_BK;
 
@@ -302,151 +306,171 @@ extern "C" void flreid_conv3x3_img_fwd(const void* X, const void* W, void* Y,
 }
 
 // ---------------------------------------------------------------------------
-// weight transpose-flip for dgrad: wT[c][r][s][k] = w[k][2-r][2-s][c]
+// 64×64 LDS-tiled bf16 transpose (wgrad operand prep): OUT[n][m] = IN[m][n]
 // ---------------------------------------------------------------------------
 
-__global__ __launch_bounds__(256) void conv3x3_wflip_kernel(
-    const __hip_bfloat16* __restrict__ W, __hip_bfloat16* __restrict__ WT,
-    int C, int K) {
-  // one thread per output element, writes coalesced over k
-  const int64_t total = (int64_t)C * 9 * K;
-  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < total;
-       i += (int64_t)gridDim.x * 256) {
-    const int k = (int)(i % K);
-    const int64_t rest = i / K;
-    const int tap = (int)(rest % 9);
-    const int c = (int)(rest / 9);
-    const int r = tap / 3, s = tap % 3;
-    WT[i] = W[((int64_t)k * 9 + (2 - r) * 3 + (2 - s)) * C + c];
+__global__ __launch_bounds__(256) void transpose_bf16_kernel(
+    const __hip_bfloat16* __restrict__ IN, __hip_bfloat16* __restrict__ OUT,
+    int64_t M, int64_t N) {
+  __shared__ __hip_bfloat16 tile[64][72];   // 8-elem pad
+  const int64_t m0 = (int64_t)blockIdx.x * 64;
+  const int64_t n0 = (int64_t)blockIdx.y * 64;
+  const int tid = threadIdx.x;
+  const int tr = tid >> 3;          // 0..31
+  const int tc8 = (tid & 7) * 8;    // 16-B column chunk
+  for (int rr = 0; rr < 64; rr += 32) {
+    const int64_t m = m0 + tr + rr;
+    u16x8 v = {};
+    if (m < M && n0 + tc8 < N) {
+      v = *(const u16x8*)(IN + m * N + n0 + tc8);
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      tile[tr + rr][tc8 + e] = *((const __hip_bfloat16*)&v + e);
+    }
+  }
+  __syncthreads();
+  for (int rr = 0; rr < 64; rr += 32) {
+    const int64_t n = n0 + tr + rr;
+    if (n >= N) continue;
+    u16x8 v;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      *((__hip_bfloat16*)&v + e) = tile[tc8 + e][tr + rr];
+    }
+    if (m0 + tc8 < M) {
+      *(u16x8*)(OUT + n * M + m0 + tc8) = v;
+    }
   }
 }
 
-extern "C" void flreid_conv3x3_wflip(const void* W, void* WT, int C, int K,
-                                     hipStream_t stream) {
-  const int64_t total = (int64_t)C * 9 * K;
-  const int blocks = (int)std::min<int64_t>((total + 255) / 256, 4096);
-  hipLaunchKernelGGL(conv3x3_wflip_kernel, dim3(blocks), dim3(256), 0, stream,
-                     (const __hip_bfloat16*)W, (__hip_bfloat16*)WT, C, K);
+extern "C" void flreid_transpose_bf16(const void* IN, void* OUT, int64_t M,
+                                      int64_t N, hipStream_t stream) {
+  dim3 grid((unsigned)((M + 63) / 64), (unsigned)((N + 63) / 64));
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream,
+                     (const __hip_bfloat16*)IN, (__hip_bfloat16*)OUT, M, N);
   HIP_CHECK(hipGetLastError());
 }
 
 // ---------------------------------------------------------------------------
-// wgrad: dθ[k][r][s][c] = Σ_m dy[m][k] · x[shift_rs(m)][c]
+// wgrad v3: dθ[k][r][s][c] = Σ_m dyT[k][m] · xT[c][shift_rs(m)]
 // ---------------------------------------------------------------------------
-// Per-tap M-reduction GEMM: A = dyᵀ (k rows), B = xᵀ (c rows), reduction
-// over flattened NHWC rows m in 32-row chunks.  Both tiles staged
-// TRANSPOSED into LDS (rows = out dims, cols = m, permuted for b128
-// fragments).  The tap's x-shift and border mask are applied at the
-// staging loads.  Output fp32, each block owns its (k-tile, c-tile, tap)
-// exclusively — no atomics.
+// Operates on PRE-TRANSPOSED operands (transpose_bf16 above): both LDS
+// panels stage coalesced 16-B row chunks with vectorized b64 writes — the
+// in-kernel transposed b16 scatter of the first version measured 75 LDS
+// bank-conflict cycles per MFMA (PMC) and is gone.  The tap's m-shift and
+// border mask are applied at the xT staging (one (n,h,w) decode per 8-m
+// chunk, incremental carry within it).
 
 constexpr int WG_BK = 64;    // k rows per block
-constexpr int WG_BC = 128;   // c cols per block
-constexpr int WG_BM = 64;    // m rows per iteration (two 32-m MFMA halves)
-constexpr int WG_MS = 80;    // padded m-stride (elems): 160 B = 40-dword
-                             // bank stride = 8*odd -> conflict-free b128
-constexpr int WG_SPLITM = 2; // M split across blocks (atomic accumulate)
-
-__device__ __forceinline__ int wg_pm(int m) {
-  // per-32 fragment permutation, two halves side by side
-  return (m >> 5) * 32 + ((m & 15) >> 2) * 8 + ((m >> 4) & 1) * 4 + (m & 3);
-}
+constexpr int WG_BC = 128;   // c rows per block
+constexpr int WG_BM = 64;    // m per iteration (two 32-m MFMA halves)
+constexpr int WG_MS = 80;    // LDS m-stride (elems)
+constexpr int WG_SPLITM = 2;
 
 __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
-    const __hip_bfloat16* __restrict__ DY, const __hip_bfloat16* __restrict__ X,
+    const __hip_bfloat16* __restrict__ DYT, const __hip_bfloat16* __restrict__ XT,
     float* __restrict__ DW, int NB, int H, int Wd, int C, int K) {
   __shared__ __hip_bfloat16 ldy[2][WG_BK * WG_MS];
   __shared__ __hip_bfloat16 lxc[2][WG_BC * WG_MS];
 
-  const int kb = blockIdx.x;          // k-block first: XCD dy-affinity
+  const int kb = blockIdx.x;          // k-block first: XCD dyT-affinity
   const int cb = blockIdx.y;
   const int tap = blockIdx.z % 9;
-  const int sp = blockIdx.z / 9;      // M-split index
+  const int sp = blockIdx.z / 9;
   const int r = tap / 3 - 1, s = tap % 3 - 1;
   const int k0 = kb * WG_BK, c0 = cb * WG_BC;
   const int HW = H * Wd;
-  const int M = NB * HW;
+  const int64_t M = (int64_t)NB * HW;
+  const int shiftM = r * Wd + s;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wk = wave & 1, wc = wave >> 1;   // 2(k) x 2(c) wave grid
+  const int wk = wave & 1, wc = wave >> 1;
   const int kg = lane >> 4;
 
-  // staging slots (loads unconditional from clamped addresses; invalid
-  // slots select zero at the LDS store — guide §5 trap c)
-  const int dy_m[2] = {tid >> 3, (tid + 256) >> 3};
-  const int dy_j[2] = {tid & 7, (tid + 256) & 7};
-  int x_m[4], x_j[4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    x_m[i] = (tid + 256 * i) >> 4;
-    x_j[i] = (tid + 256 * i) & 15;
-  }
+  // A (dyT): 2 slots/thread: row = slot>>3 (0..63), j = slot&7 (8-m chunk)
+  // B (xT): 4 slots/thread: row = slot>>3 (0..127), j = slot&7
+  u16x8 ar_[2], br_[4];
+  unsigned short bm_[4];              // per-element border masks for B
 
-  u16x8 dyr[2], xr[4];
-  bool dy_v[2], x_v[4];
-
-  auto load_tile = [&](int m0) {
+  auto load_tile = [&](int64_t m0) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int m = m0 + dy_m[i];
-      dy_v[i] = m < M && (k0 + dy_j[i] * 8) < K;
-      const int64_t src = dy_v[i]
-          ? (int64_t)m * K + k0 + dy_j[i] * 8 : (int64_t)0;
-      dyr[i] = *(const u16x8*)(DY + src);
+      const int slot = tid + 256 * i;
+      const int row = slot >> 3, j = slot & 7;
+      const int64_t kk = (int64_t)(k0 + row < K ? k0 + row : 0);
+      int64_t mc = m0 + 8 * j;
+      if (mc > M - 8) mc = M > 8 ? M - 8 : 0;    // clamp (B masks zeros)
+      ar_[i] = *(const u16x8*)(DYT + kk * M + mc);
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int m = m0 + x_m[i];
-      bool ok = m < M && (c0 + x_j[i] * 8) < C;
-      int64_t src = 0;
-      if (ok) {
-        const int n = m / HW;
-        const int rem = m - n * HW;
-        const int h = rem / Wd + r, w = rem - (rem / Wd) * Wd + s;
-        ok = h >= 0 && h < H && w >= 0 && w < Wd;
-        src = ok ? ((int64_t)n * HW + h * Wd + w) * C + c0 + x_j[i] * 8
-                 : (int64_t)0;
+      const int slot = tid + 256 * i;
+      const int row = slot >> 3, j = slot & 7;
+      const int64_t cc = (int64_t)(c0 + row < C ? c0 + row : 0);
+      const int64_t mbase = m0 + 8 * j;
+      int64_t mc = mbase + shiftM;
+      if (mc < 0) mc = 0;
+      if (mc > M - 8) mc = M > 8 ? M - 8 : 0;
+      br_[i] = *(const u16x8*)(XT + cc * M + mc);
+      // border mask: decode (n,h,w) of the first m, carry within the chunk
+      unsigned short mask = 0;
+      int mm = (int)(mbase < M ? mbase : M - 1);
+      int n = mm / HW;
+      int rem = mm - n * HW;
+      int h = rem / Wd;
+      int w = rem - h * Wd;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const bool ok = (mbase + e < M) && (h + r >= 0) && (h + r < H) &&
+                        (w + s >= 0) && (w + s < Wd) &&
+                        (mbase + e + shiftM >= 0);
+        mask |= (ok ? 1 : 0) << e;
+        if (++w == Wd) { w = 0; if (++h == H) h = 0; }
       }
-      x_v[i] = ok;
-      xr[i] = *(const u16x8*)(X + src);
+      bm_[i] = mask;
     }
   };
 
   auto store_tile = [&](int buf) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int pm = wg_pm(dy_m[i]);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        ldy[buf][(dy_j[i] * 8 + e) * WG_MS + pm] =
-            dy_v[i] ? *((const __hip_bfloat16*)&dyr[i] + e)
-                    : __float2bfloat16(0.f);
-      }
+      const int slot = tid + 256 * i;
+      const int row = slot >> 3, j = slot & 7;
+      const int pm = (j >> 2) * 32 + ci_pbase(j & 3);
+      __hip_bfloat16* dst = &ldy[buf][row * WG_MS + pm];
+      const uint64_t zero_row = (k0 + row < K) ? ~0ull : 0ull;
+      *(uint64_t*)dst = *(const uint64_t*)&ar_[i] & zero_row;
+      *(uint64_t*)(dst + 8) = *(((const uint64_t*)&ar_[i]) + 1) & zero_row;
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const int pm = wg_pm(x_m[i]);
+      const int slot = tid + 256 * i;
+      const int row = slot >> 3, j = slot & 7;
+      const int pm = (j >> 2) * 32 + ci_pbase(j & 3);
+      u16x8 v = br_[i];
+      const unsigned short mask = bm_[i];
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        lxc[buf][(x_j[i] * 8 + e) * WG_MS + pm] =
-            x_v[i] ? *((const __hip_bfloat16*)&xr[i] + e)
-                   : __float2bfloat16(0.f);
+        if (!((mask >> e) & 1)) *((__hip_bfloat16*)&v + e) = __float2bfloat16(0.f);
       }
+      __hip_bfloat16* dst = &lxc[buf][row * WG_MS + pm];
+      *(uint64_t*)dst = *(const uint64_t*)&v;
+      *(uint64_t*)(dst + 8) = *(((const uint64_t*)&v) + 1);
     }
   };
 
   if32x4 acc[2][4] = {};
 
-  // this split's chunk range
-  const int NC = (M + WG_BM - 1) / WG_BM;
+  const int NC = (int)((M + WG_BM - 1) / WG_BM);
   const int per = (NC + WG_SPLITM - 1) / WG_SPLITM;
   const int ch0 = sp * per;
   const int ch1 = min(NC, ch0 + per);
+  float* dst_part = DW + (int64_t)sp * K * 9 * C;
   if (ch0 >= ch1) {
     // empty split: its partial tile must still be zero-filled
-    float* dst = DW + (int64_t)sp * K * 9 * C;
 #pragma unroll
     for (int fk = 0; fk < 2; ++fk)
 #pragma unroll
@@ -455,26 +479,26 @@ __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
         for (int reg = 0; reg < 4; ++reg) {
           const int k = k0 + wk * 32 + fk * 16 + (lane >> 4) * 4 + reg;
           const int c = c0 + wc * 64 + fc * 16 + (lane & 15);
-          if (k < K && c < C) dst[((int64_t)k * 9 + tap) * C + c] = 0.f;
+          if (k < K && c < C) dst_part[((int64_t)k * 9 + tap) * C + c] = 0.f;
         }
     return;
   }
 
-  load_tile(ch0 * WG_BM);
+  load_tile((int64_t)ch0 * WG_BM);
   store_tile(0);
-  if (ch0 + 1 < ch1) load_tile((ch0 + 1) * WG_BM);
+  if (ch0 + 1 < ch1) load_tile((int64_t)(ch0 + 1) * WG_BM);
   __syncthreads();
 
   for (int it = ch0; it < ch1; ++it) {
     const int buf = (it - ch0) & 1;
     if (it + 1 < ch1) {
       store_tile(buf ^ 1);
-      if (it + 2 < ch1) load_tile((it + 2) * WG_BM);
+      if (it + 2 < ch1) load_tile((int64_t)(it + 2) * WG_BM);
     }
     const __hip_bfloat16* db = ldy[buf];
     const __hip_bfloat16* xb = lxc[buf];
 #pragma unroll
-    for (int hh = 0; hh < 2; ++hh) {       // two 32-m halves per tile
+    for (int hh = 0; hh < 2; ++hh) {
       ibf16x8 afrag[2];
 #pragma unroll
       for (int fk = 0; fk < 2; ++fk) {
@@ -495,10 +519,6 @@ __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
     __syncthreads();
   }
 
-  // D rows = k (A), cols = c (B).  Each M split writes its OWN partial
-  // buffer with plain stores (4.7M fp32 atomicAdds measured ~0.4 ms on the
-  // bench shape — the wrapper sums the two 9.4 MB partials in ~3 µs)
-  float* dst = DW + (int64_t)sp * K * 9 * C;
 #pragma unroll
   for (int fk = 0; fk < 2; ++fk) {
 #pragma unroll
@@ -508,24 +528,25 @@ __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
         const int k = k0 + wk * 32 + fk * 16 + (lane >> 4) * 4 + reg;
         const int c = c0 + wc * 64 + fc * 16 + (lane & 15);
         if (k < K && c < C) {
-          dst[((int64_t)k * 9 + tap) * C + c] = acc[fk][fc][reg];
+          dst_part[((int64_t)k * 9 + tap) * C + c] = acc[fk][fc][reg];
         }
       }
     }
   }
 }
 
-extern "C" void flreid_conv3x3_wgrad(const void* DY, const void* X, float* DW,
-                                     int NB, int H, int Wd, int C, int K,
-                                     hipStream_t stream) {
+extern "C" void flreid_conv3x3_wgrad(const void* DYT, const void* XT,
+                                     float* DW, int NB, int H, int Wd, int C,
+                                     int K, hipStream_t stream) {
   if (C % 8 || K % 8) {
     throw std::runtime_error("conv3x3_wgrad: C%8 or K%8 != 0");
   }
-  // DW = [WG_SPLITM][K][9][C] fp32 partials; caller sums over splits
+  // DW = [WG_SPLITM][K][9][C] fp32 partials; caller sums over splits.
+  // DYT/XT are the TRANSPOSED [K][M] / [C][M] operands.
   dim3 grid((K + WG_BK - 1) / WG_BK, (C + WG_BC - 1) / WG_BC, 9 * WG_SPLITM);
   hipLaunchKernelGGL(conv3x3_wgrad_kernel, grid, dim3(256), 0, stream,
-                     (const __hip_bfloat16*)DY, (const __hip_bfloat16*)X, DW,
-                     NB, H, Wd, C, K);
+                     (const __hip_bfloat16*)DYT, (const __hip_bfloat16*)XT,
+                     DW, NB, H, Wd, C, K);
   HIP_CHECK(hipGetLastError());
 }
 

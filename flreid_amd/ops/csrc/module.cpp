@@ -65,8 +65,8 @@ extern "C" void flreid_compose2(const void*, const float*, const void*,
                                 hipStream_t);
 extern "C" void flreid_conv3x3_img_fwd(const void*, const void*, void*, int,
                                        int, int, int, int, hipStream_t);
-extern "C" void flreid_conv3x3_wflip(const void*, void*, int, int,
-                                     hipStream_t);
+extern "C" void flreid_transpose_bf16(const void*, void*, int64_t, int64_t,
+                                      hipStream_t);
 extern "C" void flreid_conv3x3_tile(const void*, const float*, const void*,
                                     void*, int, int, int, int, hipStream_t);
 extern "C" void flreid_conv3x3_img_fwd_ldsw(const void*, const void*, void*,
@@ -262,10 +262,11 @@ PYBIND11_MODULE(_flreid_hip, m) {
                                          as_stream(stream));
         });
 
-  m.def("conv3x3_wflip",
-        [](uintptr_t w, uintptr_t wt, int C, int K, uintptr_t stream) {
-          flreid::flreid_conv3x3_wflip((const void*)w, (void*)wt, C, K,
-                                       as_stream(stream));
+  m.def("transpose_bf16",
+        [](uintptr_t in, uintptr_t out, int64_t M, int64_t N,
+           uintptr_t stream) {
+          flreid::flreid_transpose_bf16((const void*)in, (void*)out, M, N,
+                                        as_stream(stream));
         });
 
   m.def("conv3x3_img_fwd_ldsw",

@@ -1,0 +1,37 @@
+"""Throughput regression guards (VERDICT round-1 item 8).
+
+Encodes the bench ladder as a test: a handful of flagship federated rounds
+must stay above a floor set at 0.8x the round-1 driver-recorded number
+(4209 img/s, BENCH_r01.json), so kernel work cannot silently regress the
+round.  Box-to-box thermal variance measured in round 1 was about +-10%
+(worst observed throttled box: 3503 img/s — still above this floor).
+"""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+FLOOR_IMGS_PER_SEC = 3400.0   # 0.8 x 4250 (round-1 fresh-box band)
+
+
+@pytest.mark.timeout(600)
+def test_fedstil_resnet50_round_throughput_floor():
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"),
+         "--steps", "6", "--warmup", "2"],
+        capture_output=True, text=True, timeout=540, env=env, cwd=repo)
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert lines, f"no bench JSON line\n{out.stdout[-2000:]}\n{out.stderr[-2000:]}"
+    result = json.loads(lines[-1])
+    assert result["value"] >= FLOOR_IMGS_PER_SEC, \
+        (f"flagship round throughput {result['value']} img/s fell below the "
+         f"{FLOOR_IMGS_PER_SEC} floor (round-1 band 4176-4297; "
+         f"ms_per_step={result['ms_per_step']})")
