@@ -66,13 +66,13 @@ def run_shape(n, c, h, w, k):
     def ours_wgrad():
         st = torch.cuda.current_stream().cuda_stream
         m_rows = n * h * w
-        dyt = torch.empty(k, m_rows, device="cuda", dtype=torch.bfloat16)
-        xt = torch.empty(c, m_rows, device="cuda", dtype=torch.bfloat16)
-        ext.transpose_bf16(dy.data_ptr(), dyt.data_ptr(), m_rows, k, st)
-        ext.transpose_bf16(x.data_ptr(), xt.data_ptr(), m_rows, c, st)
+        dyt = torch.empty(k * m_rows + 256, device="cuda", dtype=torch.bfloat16)
+        xt = torch.empty(c * m_rows + 256, device="cuda", dtype=torch.bfloat16)
+        ext.transpose_bf16(dy.data_ptr(), dyt.data_ptr() + 256, m_rows, k, st)
+        ext.transpose_bf16(x.data_ptr(), xt.data_ptr() + 256, m_rows, c, st)
         part = torch.empty(2, k * 9 * c, device="cuda", dtype=torch.float32)
-        ext.conv3x3_wgrad(dyt.data_ptr(), xt.data_ptr(), part.data_ptr(),
-                          n, h, w, c, k, st)
+        ext.conv3x3_wgrad(dyt.data_ptr() + 256, xt.data_ptr() + 256,
+                          part.data_ptr(), n, h, w, c, k, st)
         return (part[0] + part[1]).view(k, 3, 3, c).permute(0, 3, 1, 2)
 
     def ours_fwd_ldsw():

@@ -913,17 +913,30 @@ def conv_theta_tile(ext, w: torch.Tensor, atten: Optional[torch.Tensor],
     return out
 
 
-class _Conv3x3Fn(torch.autograd.Function):
-    """Hand-written CDNA4 3×3 s1p1 conv (conv3x3_img.hip) with the FedSTIL
-    composition fused into the weight production (K1).
+def _conv_impl() -> str:
+    """Per-op conv engine policy.  'auto' (default) picks the MEASURED
+    winner per op (profiles/README.md K1 table): the library conv
+    currently beats the hand kernels on the ReID shapes (CK fwd 529 TF vs
+    hand 428; wgrad 380 vs 103), so auto routes the conv math to the
+    library while keeping the fused one-pass θ production (compose2 to
+    bf16 — no fp32 θ, no autocast cast) and the fp32 weight-grad path.
+    'hand' forces the hand-written kernels everywhere (the A/B switch the
+    probe and profiles use); 'lib' forces the library."""
+    return os.environ.get("FLREID_CONV_IMPL", "auto")
 
-    fwd:   θ_bf16 = compose2(gw, atten, aw) (or a cast of a plain weight);
-           y = conv3x3_img_fwd(x, θ) — one-image-per-block halo kernel.
-    dgrad: dx = conv3x3_img_fwd(dy, tile(θ, flipped)) — the same kernel on
-           the flip-transposed weight tile.
-    wgrad: dθ = conv3x3_wgrad(dy, x), fp32 accumulate/output; for the
-           adaptive layer dθ IS d(adaptive_weight) (identity composition),
-           so no bf16 round-trip on the weight gradient.
+
+class _Conv3x3Fn(torch.autograd.Function):
+    """3×3 s1p1 conv with the FedSTIL composition fused into the weight
+    production (K1).
+
+    fwd:   θ_bf16 tile/tensor from ONE compose2/tile pass over
+           (gw, atten, aw); conv = hand halo kernel (conv3x3_img*.hip) or
+           the library on θ, by policy (_conv_impl).
+    dgrad: hand = the same kernel on the flip-transposed θ tile;
+           lib = conv2d_input on θ.
+    wgrad: hand = transpose-first MFMA reduction (fp32 out);
+           lib = conv2d_weight (bf16, cast once).  For the adaptive layer
+           dθ IS d(adaptive_weight) (identity composition).
     Only frozen gw/atten are supported (FedSTIL trains aw alone;
     plain nn.Conv2d passes the weight in slot 1)."""
 
@@ -933,11 +946,16 @@ class _Conv3x3Fn(torch.autograd.Function):
         x_bf = _cl(x.detach().to(torch.bfloat16))
         n, c, h, w = x_bf.shape
         k = weight.shape[0]
-        theta_tile = conv_theta_tile(ext, weight, atten, aw, mode=0)
-        y = torch.empty(n, k, h, w, device=x.device, dtype=torch.bfloat16,
-                        memory_format=torch.channels_last)
-        ext.conv3x3_img_fwd(x_bf.data_ptr(), theta_tile.data_ptr(),
-                            y.data_ptr(), n, h, w, c, k, _stream())
+        hand = _conv_impl() == "hand"
+        if hand:
+            theta = compose_theta_bf16(ext, weight, atten, aw)  # plain CL θ
+            y = torch.empty(n, k, h, w, device=x.device, dtype=torch.bfloat16,
+                            memory_format=torch.channels_last)
+            ext.conv3x3_img_fwd_ldsw(x_bf.data_ptr(), theta.data_ptr(),
+                                     y.data_ptr(), n, h, w, c, k, _stream())
+        else:
+            theta = compose_theta_bf16(ext, weight, atten, aw)
+            y = torch.nn.functional.conv2d(x_bf, theta, padding=1)
         ctx.save_for_backward(x_bf, weight, atten, aw)
         ctx.x_dtype = x.dtype
         ctx.w_dtype = weight.dtype
@@ -949,38 +967,51 @@ class _Conv3x3Fn(torch.autograd.Function):
         x_bf, weight, atten, aw = ctx.saved_tensors
         n, c, h, w = x_bf.shape
         k = weight.shape[0]
+        hand = _conv_impl() == "hand"
         dy_bf = _cl(dy.to(torch.bfloat16))
+        theta = None
         dx = None
         if ctx.needs_input_grad[0]:
-            # dgrad = the same kernel on (dy, flipped-transposed θ tile)
-            wt_tile = conv_theta_tile(ext, weight, atten, aw, mode=1)
-            dx = torch.empty(n, c, h, w, device=dy.device,
-                             dtype=torch.bfloat16,
-                             memory_format=torch.channels_last)
-            ext.conv3x3_img_fwd(dy_bf.data_ptr(), wt_tile.data_ptr(),
-                                dx.data_ptr(), n, h, w, k, c, _stream())
+            if hand:
+                # dgrad = the same kernel on the flip-transposed θ tile
+                wt_tile = conv_theta_tile(ext, weight, atten, aw, mode=1)
+                dx = torch.empty(n, c, h, w, device=dy.device,
+                                 dtype=torch.bfloat16,
+                                 memory_format=torch.channels_last)
+                ext.conv3x3_img_fwd(dy_bf.data_ptr(), wt_tile.data_ptr(),
+                                    dx.data_ptr(), n, h, w, k, c, _stream())
+            else:
+                theta = compose_theta_bf16(ext, weight, atten, aw)
+                dx = torch.nn.grad.conv2d_input((n, c, h, w), theta, dy_bf,
+                                                padding=1)
             if ctx.x_dtype != torch.bfloat16:
                 dx = dx.to(ctx.x_dtype)
         d_weight = d_aw = None
         if ctx.needs_input_grad[1] or ctx.needs_input_grad[3]:
-            # wgrad on pre-transposed operands (the in-kernel transposed
-            # scatter measured 75 LDS-conflict cycles/MFMA); two exclusive
-            # M-split partials summed here
-            m_rows = n * h * w
-            dyt = torch.empty(k, m_rows, device=dy.device,
-                              dtype=torch.bfloat16)
-            xt = torch.empty(c, m_rows, device=dy.device,
-                             dtype=torch.bfloat16)
-            ext.transpose_bf16(dy_bf.data_ptr(), dyt.data_ptr(), m_rows, k,
-                               _stream())
-            ext.transpose_bf16(x_bf.data_ptr(), xt.data_ptr(), m_rows, c,
-                               _stream())
-            part = torch.empty(2, k * 9 * c, device=dy.device,
-                               dtype=torch.float32)
-            ext.conv3x3_wgrad(dyt.data_ptr(), xt.data_ptr(),
-                              part.data_ptr(), n, h, w, c, k, _stream())
-            # [K][9][C] flat == the channels-last [K,C,3,3] physical layout
-            dw = (part[0] + part[1]).view(k, 3, 3, c).permute(0, 3, 1, 2)
+            if hand:
+                # wgrad on pre-transposed operands (the in-kernel
+                # transposed scatter measured 75 LDS-conflict cycles/MFMA);
+                # two exclusive M-split partials summed here
+                m_rows = n * h * w
+                # 128-elem pads both ends: the wgrad kernel loads
+                # shifted/tail chunks unclamped (OOR lanes zero-masked)
+                dyt = torch.empty(k * m_rows + 256, device=dy.device,
+                                  dtype=torch.bfloat16)
+                xt = torch.empty(c * m_rows + 256, device=dy.device,
+                                 dtype=torch.bfloat16)
+                ext.transpose_bf16(dy_bf.data_ptr(), dyt.data_ptr() + 256,
+                                   m_rows, k, _stream())
+                ext.transpose_bf16(x_bf.data_ptr(), xt.data_ptr() + 256,
+                                   m_rows, c, _stream())
+                part = torch.empty(2, k * 9 * c, device=dy.device,
+                                   dtype=torch.float32)
+                ext.conv3x3_wgrad(dyt.data_ptr() + 256, xt.data_ptr() + 256,
+                                  part.data_ptr(), n, h, w, c, k, _stream())
+                # [K][9][C] flat == channels-last [K,C,3,3] physical layout
+                dw = (part[0] + part[1]).view(k, 3, 3, c).permute(0, 3, 1, 2)
+            else:
+                dw = torch.nn.grad.conv2d_weight(
+                    x_bf, (k, c, 3, 3), dy_bf, padding=1).to(torch.float32)
             if ctx.needs_input_grad[3]:
                 d_aw = dw              # identity composition, fp32 direct
             else:

@@ -395,14 +395,17 @@ __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
   u16x8 ar_[2], br_[4];
   unsigned short bm_[4];              // per-element border masks for B
 
+  // DYT/XT carry ≥128-elem pads on both ends (the wrapper allocates them),
+  // so shifted/tail chunk addresses are loaded UNCLAMPED — a clamp would
+  // misalign the chunk's element<->m correspondence.  Out-of-range lanes
+  // read pad/neighbour-row bytes and are zero-masked at the LDS store.
   auto load_tile = [&](int64_t m0) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       const int slot = tid + 256 * i;
       const int row = slot >> 3, j = slot & 7;
       const int64_t kk = (int64_t)(k0 + row < K ? k0 + row : 0);
-      int64_t mc = m0 + 8 * j;
-      if (mc > M - 8) mc = M > 8 ? M - 8 : 0;    // clamp (B masks zeros)
+      const int64_t mc = m0 + 8 * j;
       ar_[i] = *(const u16x8*)(DYT + kk * M + mc);
     }
 #pragma unroll
@@ -411,9 +414,7 @@ __global__ __launch_bounds__(256) void conv3x3_wgrad_kernel(
       const int row = slot >> 3, j = slot & 7;
       const int64_t cc = (int64_t)(c0 + row < C ? c0 + row : 0);
       const int64_t mbase = m0 + 8 * j;
-      int64_t mc = mbase + shiftM;
-      if (mc < 0) mc = 0;
-      if (mc > M - 8) mc = M > 8 ? M - 8 : 0;
+      const int64_t mc = mbase + shiftM;
       br_[i] = *(const u16x8*)(XT + cc * M + mc);
       // border mask: decode (n,h,w) of the first m, carry within the chunk
       unsigned short mask = 0;

@@ -14,7 +14,10 @@ using ibf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using if32x4 = __attribute__((ext_vector_type(4))) float;
 using u16x8 = __attribute__((ext_vector_type(8))) unsigned short;
 
-constexpr int CW_CS = 40;        // padded per-cell c-stride (elems)
+constexpr int CW_CS = 48;        // padded per-cell c-stride: 24-dword
+                                 // bank stride = 8*odd -> conflict-free
+                                 // ds_read_b128 lane groups (PMC: 9.7
+                                 // conflict-cycles/MFMA at stride 40)
 constexpr int CW_CELLS = 18 * 10;  // max (H+2)*(W+2)
 constexpr int CW_BK = 32;        // input-channel tile
 constexpr int CW_BN = 64;        // output channels per block

@@ -518,9 +518,11 @@ def _cl4(t):
     return t.to(memory_format=torch.channels_last)
 
 
-def test_conv3x3_img_fwd_matches_torch():
-    """K1 specialized one-image-per-block halo kernel (conv3x3_img.hip)
-    fwd vs the fp32 library conv on bf16-rounded inputs."""
+@pytest.mark.parametrize("impl", ["auto", "hand"])
+def test_conv3x3_img_fwd_matches_torch(impl, monkeypatch):
+    """K1 conv path (conv3x3_img*.hip / fused-compose library route) fwd
+    vs the fp32 library conv on bf16-rounded inputs."""
+    monkeypatch.setenv("FLREID_CONV_IMPL", impl)
     torch.manual_seed(1)
     for (n, c, h, w, k) in ((4, 32, 16, 8, 64), (3, 64, 8, 4, 32),
                             (8, 512, 16, 8, 512), (2, 256, 16, 8, 512)):
@@ -536,9 +538,12 @@ def test_conv3x3_img_fwd_matches_torch():
             (n, c, h, w, k, (out.float() - expected).abs().max())
 
 
-def test_conv3x3_img_plain_bwd_matches_torch():
-    """dgrad (flip-transpose + the same fwd kernel) and wgrad (per-tap
-    M-reduction MFMA kernel, fp32 out) vs torch autograd in fp32."""
+@pytest.mark.parametrize("impl", ["auto", "hand"])
+def test_conv3x3_img_plain_bwd_matches_torch(impl, monkeypatch):
+    """dgrad (flip-transposed tile + the fwd kernel / conv2d_input) and
+    wgrad (transpose-first MFMA reduction / conv2d_weight) vs torch
+    autograd in fp32."""
+    monkeypatch.setenv("FLREID_CONV_IMPL", impl)
     torch.manual_seed(2)
     n, c, h, w, k = 4, 64, 16, 8, 64
     x = _cl4(torch.randn(n, c, h, w, device="cuda").bfloat16())
@@ -562,9 +567,11 @@ def test_conv3x3_img_plain_bwd_matches_torch():
         (x1.grad.float() - x2.grad).abs().max()
 
 
-def test_conv3x3_img_adaptive_fwd_bwd():
-    """Composed path: θ = atten⊙gw + aw fused into the conv weight fetch;
-    d(aw) comes straight out of wgrad in fp32 (identity composition)."""
+@pytest.mark.parametrize("impl", ["auto", "hand"])
+def test_conv3x3_img_adaptive_fwd_bwd(impl, monkeypatch):
+    """Composed path: θ = atten⊙gw + aw fused into the conv weight
+    production; d(aw) arrives in fp32 (identity composition)."""
+    monkeypatch.setenv("FLREID_CONV_IMPL", impl)
     torch.manual_seed(3)
     n, c, h, w, k = 2, 32, 16, 8, 32
     x = _cl4(torch.randn(n, c, h, w, device="cuda").bfloat16())
