@@ -34,10 +34,14 @@ def needs_rebuild() -> bool:
 
 
 def build(force: bool = False, arch: str = "gfx950", verbose: bool = True,
-          debug: bool = False) -> str:
-    """debug=True (or --debug / FLREID_BUILD_DEBUG=1): -g -O1 plus host
-    address-sanitizer — the kernel-debug build for rocgdb / sanitizer runs
-    (SURVEY.md §5.2); release build is the default."""
+          debug: bool = False, asan: bool = False) -> str:
+    """debug=True (or --debug / FLREID_BUILD_DEBUG=1): -g -O1 — the
+    kernel-debug build for rocgdb / serialized-fault runs.
+    asan=True (or --asan / FLREID_BUILD_ASAN=1): additionally instruments
+    the HOST side with AddressSanitizer; run python with
+    LD_PRELOAD=$(/opt/rocm/lib/llvm/bin/clang
+    -print-file-name=libclang_rt.asan-x86_64.so) ASAN_OPTIONS=detect_leaks=0.
+    Release build is the default."""
     if not force and not needs_rebuild():
         if verbose:
             print(f"[flreid build] up to date: {OUT}")
@@ -45,9 +49,15 @@ def build(force: bool = False, arch: str = "gfx950", verbose: bool = True,
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
     includes = [sysconfig.get_paths()["include"]] + _pybind11_includes()
     debug = debug or os.environ.get("FLREID_BUILD_DEBUG", "0") == "1"
-    opt = ["-g", "-O1"] if debug else ["-O3"]
+    asan = asan or os.environ.get("FLREID_BUILD_ASAN", "0") == "1"
+    opt = ["-g", "-O1"] if (debug or asan) else ["-O3"]
+    extra = []
+    if asan:
+        # host-side ASan only: device code stays uninstrumented (xnack-free)
+        extra = ["-fsanitize=address", "-shared-libsan",
+                 "-Xarch_device", "-fno-sanitize=all"]
     cmd = [
-        hipcc, f"--offload-arch={arch}", *opt, "-std=c++17", "-fPIC",
+        hipcc, f"--offload-arch={arch}", *opt, *extra, "-std=c++17", "-fPIC",
         "-shared", "-fvisibility=hidden",
         *[f"-I{p}" for p in includes],
         *[os.path.join(CSRC, s) for s in SOURCES],
@@ -60,5 +70,5 @@ def build(force: bool = False, arch: str = "gfx950", verbose: bool = True,
 
 
 if __name__ == "__main__":
-    build(force="--force" in sys.argv or "--debug" in sys.argv,
-          debug="--debug" in sys.argv)
+    build(force=any(f in sys.argv for f in ("--force", "--debug", "--asan")),
+          debug="--debug" in sys.argv, asan="--asan" in sys.argv)
