@@ -31,7 +31,11 @@ def build_cfg(method: str, out_dir: str, args):
     common = {
         "datasets_dir": (f"synthetic://ids={args.ids},train={args.imgs},"
                          f"query=2,gallery=4,hw=128x64,idspace=4096"),
-        "checkpoints_dir": os.path.join(out_dir, f"ckpts-{method}"),
+        # ckpts are run-local scratch (the lifelong task-switch reloads need
+        # them) — keep them OUT of the result dir: ~100 MB of model states
+        # would blow the gpurun_out copy-back budget
+        "checkpoints_dir": os.path.join(
+            os.environ.get("TMPDIR", "/tmp"), f"fid_ckpts-{method}"),
         "logs_dir": out_dir,
         "parallel": 1,
         "device": ["cpu" if args.cpu else "cuda:0"],
