@@ -782,8 +782,9 @@ def bn_train_2d(x: torch.Tensor, bn,
     """Fused TRAINING-mode BatchNorm for the head-epoch regime.  Applies to
     CUDA channels-last 4-D tensors (BatchNorm2d) and contiguous 2-D tensors
     (BatchNorm1d — the BNNeck bottleneck) with C % 64 == 0 and a small row
-    count (M = N·H·W ≤ 4096 — the cached-prototype batches; large-M
-    full-image training stays on MIOpen's multi-block reduction).  The
+    count (M = N·H·W ≤ 16384 — covers the cached-prototype batches AND
+    batch-64 layer-4 full-image fine-tuning at 16×8 spatial; very large M
+    stays on MIOpen's multi-block reduction).  The
     num_batches_tracked increment is fused into the forward kernel.  Returns
     None when the fused path does not apply."""
     if os.environ.get("FLREID_NO_FUSED_BN", "0") == "1":
@@ -801,7 +802,7 @@ def bn_train_2d(x: torch.Tensor, bn,
         return None
     c = x.shape[1]
     m = x.numel() // c
-    if c % 64 != 0 or m < 2 or m > 4096:
+    if c % 64 != 0 or m < 2 or m > 16384:
         return None
     if bn.momentum is None:  # cumulative-average mode: keep torch semantics
         return None

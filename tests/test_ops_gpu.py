@@ -398,6 +398,32 @@ def test_bn_train_fused_declines_out_of_regime():
     assert ops.bn_train_2d(x2, bn2) is None
 
 
+def test_bn_train_fused_large_m_matches_torch():
+    """The full-image fine-tune regime (fedavg/fedprox layer-4: M = 8192
+    rows) through the fused train-BN — raised from the round-1 M<=4096
+    cap."""
+    import torch.nn as nn
+    torch.manual_seed(9)
+    bn = nn.BatchNorm2d(128).cuda().train()
+    bn_ref = nn.BatchNorm2d(128).cuda().train()
+    bn_ref.load_state_dict(bn.state_dict())
+    x = (torch.randn(64, 128, 16, 8, device="cuda")
+         .to(memory_format=torch.channels_last).requires_grad_(True))
+    x2 = x.detach().clone().requires_grad_(True)
+    y = ops.bn_train_2d(x, bn)
+    assert y is not None
+    y_ref = bn_ref(x2)
+    assert torch.allclose(y, y_ref, atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(bn.weight.grad, bn_ref.weight.grad, atol=1e-3,
+                          rtol=1e-3)
+    assert torch.allclose(bn.running_mean, bn_ref.running_mean, atol=1e-5)
+    assert torch.allclose(bn.running_var, bn_ref.running_var, atol=1e-4)
+
+
 def test_bn_train_fused_1d_matches_torch():
     """2-D input path (BNNeck BatchNorm1d) of the fused train BN."""
     import copy
