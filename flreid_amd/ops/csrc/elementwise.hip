@@ -158,12 +158,47 @@ __global__ void compose2_kernel(const Tin* __restrict__ gw,
                                 const Tin* __restrict__ aw,
                                 Tout* __restrict__ out, int64_t numel,
                                 int64_t L, int64_t inner) {
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= numel) return;
-  const float a = atten ? atten[(i / inner) % L] : 1.0f;
-  const float v = fmaf(a, load_as_float(gw, i),
-                       aw ? load_as_float(aw, i) : 0.0f);
-  store_from_float(out, i, v);
+  // 4 elements per thread (vectorized when the 4-run stays inside one
+  // atten-broadcast segment — always true for inner % 4 == 0 or L == 1):
+  // the scalar form measured 0.37 TB/s in the bench trace (launch+issue
+  // bound on the 13 per-step compositions)
+  const int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i0 >= numel) return;
+  if (i0 + 4 <= numel && (L == 1 || !atten || (inner % 4 == 0))) {
+    const float a = atten ? atten[(i0 / inner) % L] : 1.0f;
+    float4 g, w = {0.f, 0.f, 0.f, 0.f};
+    if constexpr (sizeof(Tin) == 4) {
+      g = *(const float4*)(gw + i0);
+      if (aw) w = *(const float4*)(aw + i0);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        (&g.x)[e] = load_as_float(gw, i0 + e);
+        if (aw) (&w.x)[e] = load_as_float(aw, i0 + e);
+      }
+    }
+    Tout v4[4];
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      float r = fmaf(a, (&g.x)[e], (&w.x)[e]);
+      if constexpr (sizeof(Tout) == 4) {
+        v4[e] = r;
+      } else {
+        v4[e] = __float2bfloat16(r);
+      }
+    }
+    __builtin_memcpy(out + i0, v4, sizeof(v4));
+    return;
+  }
+#pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    const int64_t i = i0 + e;
+    if (i >= numel) return;
+    const float a = atten ? atten[(i / inner) % L] : 1.0f;
+    const float v = fmaf(a, load_as_float(gw, i),
+                         aw ? load_as_float(aw, i) : 0.0f);
+    store_from_float(out, i, v);
+  }
 }
 
 extern "C" void flreid_compose2(const void* gw, const float* atten,
@@ -171,7 +206,7 @@ extern "C" void flreid_compose2(const void* gw, const float* atten,
                                 int64_t L, int64_t inner, int in_dtype,
                                 int out_dtype, hipStream_t stream) {
   constexpr int BLOCK = 256;
-  dim3 grid((unsigned)((numel + BLOCK - 1) / BLOCK)), block(BLOCK);
+  dim3 grid((unsigned)((numel + BLOCK * 4 - 1) / (BLOCK * 4))), block(BLOCK);
   if (in_dtype == kF32 && out_dtype == kBF16) {
     hipLaunchKernelGGL((compose2_kernel<float, __hip_bfloat16>), grid, block,
                        0, stream, (const float*)gw, atten, (const float*)aw,
