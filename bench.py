@@ -51,7 +51,10 @@ def _seed_miopen_db():
         return
     import shutil
     import tempfile
-    dst = os.path.join(tempfile.gettempdir(), "flreid_miopen_udb")
+    # per-rank dir: MIOpen WRITES to the user DB path, and the driver's
+    # 8-rank SCALE launch would otherwise race all ranks on one file set
+    rank = os.environ.get("RANK", "0")
+    dst = os.path.join(tempfile.gettempdir(), f"flreid_miopen_udb_r{rank}")
     os.makedirs(dst, exist_ok=True)
     for f in os.listdir(src):
         if not f.endswith(".txt"):
