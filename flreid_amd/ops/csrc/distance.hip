@@ -36,8 +36,8 @@ __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
     const float* __restrict__ aa, const float* __restrict__ bb,
     float* __restrict__ OUT, int64_t M, int64_t N, int64_t D) {
-  __shared__ float ldsA[2][BM][BK + PAD];
-  __shared__ float ldsB[2][BM][BK + PAD];
+  __shared__ float ldsA[BM][BK + PAD];
+  __shared__ float ldsB[BN][BK + PAD];
 
   // note: an XCD-aware block swizzle (guide T1) measured neutral here —
   // x-major dispatch already gives consecutive blocks a shared B panel
@@ -53,9 +53,7 @@ __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
 
   // float4 staging: 8 lanes cover one 32-float K-row; interior blocks skip
   // every bounds check and the int64 per-element address math (PMC showed
-  // guarded scalar loads made the kernel VALU-bound: 4.5 VALU/MFMA).
-  // T14 pipeline: loads for panel t+2 are issued before panel t's MFMA
-  // burst; double-buffered LDS -> ONE barrier per panel (was two).
+  // guarded scalar loads made the kernel VALU-bound: 4.5 VALU/MFMA)
   const int lc4 = (tid & 7) * 4;        // load col 0,4,...,28
   const int lr0 = tid >> 3;             // load row 0..31 (32 rows/pass)
   const bool interior = (row0 + BM <= M) && (col0 + BN <= N) && (D % 4 == 0);
@@ -63,87 +61,41 @@ __global__ __launch_bounds__(256) void pairwise_mfma_kernel(
   const int fr = lane & 31;             // fragment row/col within 32-tile
   const int fk = lane >> 5;             // fragment k (0/1)
 
-  float4 ra[4], rb[4];
-  int64_t k0_pend = 0;
-
-  auto load_regs = [&](int64_t k0) {
-    k0_pend = k0;
+  for (int64_t k0 = 0; k0 < D; k0 += BK) {
     if (interior && k0 + BK <= D) {
       const float* arow = A + (row0 + lr0) * D + k0 + lc4;
       const float* brow = B + (col0 + lr0) * D + k0 + lc4;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        ra[r] = *(const float4*)(arow + (int64_t)(r * 32) * D);
-        rb[r] = *(const float4*)(brow + (int64_t)(r * 32) * D);
+      for (int r = 0; r < BM; r += 32) {
+        const float4 va = *(const float4*)(arow + (int64_t)r * D);
+        const float4 vb = *(const float4*)(brow + (int64_t)r * D);
+        ldsA[lr0 + r][lc4 + 0] = va.x; ldsA[lr0 + r][lc4 + 1] = va.y;
+        ldsA[lr0 + r][lc4 + 2] = va.z; ldsA[lr0 + r][lc4 + 3] = va.w;
+        ldsB[lr0 + r][lc4 + 0] = vb.x; ldsB[lr0 + r][lc4 + 1] = vb.y;
+        ldsB[lr0 + r][lc4 + 2] = vb.z; ldsB[lr0 + r][lc4 + 3] = vb.w;
       }
     } else {
-      // edge panel: UNCONDITIONAL loads from clamped addresses (branching
-      // around loads serialises them — guide §5 trap c); zeroed at store
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int64_t ar = row0 + lr0 + r * 32;
-        const int64_t bc = col0 + lr0 + r * 32;
-        float av[4], bv[4];
+      for (int r = 0; r < BM; r += 32) {
+        const int lr = lr0 + r;
+        const int64_t ar = row0 + lr;
+        const int64_t bc = col0 + lr;
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
           const int64_t kk = k0 + lc4 + c;
-          const int64_t ia = (ar < M && kk < D) ? ar * D + kk : 0;
-          const int64_t ib = (bc < N && kk < D) ? bc * D + kk : 0;
-          av[c] = A[ia];
-          bv[c] = B[ib];
-        }
-        ra[r] = float4{av[0], av[1], av[2], av[3]};
-        rb[r] = float4{bv[0], bv[1], bv[2], bv[3]};
-      }
-    }
-  };
-
-  auto store_panel = [&](int buf) {
-    if (interior && k0_pend + BK <= D) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int lr = lr0 + r * 32;
-        ldsA[buf][lr][lc4 + 0] = ra[r].x; ldsA[buf][lr][lc4 + 1] = ra[r].y;
-        ldsA[buf][lr][lc4 + 2] = ra[r].z; ldsA[buf][lr][lc4 + 3] = ra[r].w;
-        ldsB[buf][lr][lc4 + 0] = rb[r].x; ldsB[buf][lr][lc4 + 1] = rb[r].y;
-        ldsB[buf][lr][lc4 + 2] = rb[r].z; ldsB[buf][lr][lc4 + 3] = rb[r].w;
-      }
-    } else {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int lr = lr0 + r * 32;
-        const int64_t ar = row0 + lr;
-        const int64_t bc = col0 + lr;
-        const float* av = (const float*)&ra[r];
-        const float* bv = (const float*)&rb[r];
-#pragma unroll
-        for (int c = 0; c < 4; ++c) {
-          const int64_t kk = k0_pend + lc4 + c;
-          ldsA[buf][lr][lc4 + c] = (ar < M && kk < D) ? av[c] : 0.0f;
-          ldsB[buf][lr][lc4 + c] = (bc < N && kk < D) ? bv[c] : 0.0f;
+          ldsA[lr][lc4 + c] = (ar < M && kk < D) ? A[ar * D + kk] : 0.0f;
+          ldsB[lr][lc4 + c] = (bc < N && kk < D) ? B[bc * D + kk] : 0.0f;
         }
       }
     }
-  };
+    __syncthreads();
 
-  const int64_t NP = (D + BK - 1) / BK;
-  load_regs(0);
-  store_panel(0);
-  if (NP > 1) load_regs(BK);
-  __syncthreads();
-
-  for (int64_t t = 0; t < NP; ++t) {
-    const int buf = (int)(t & 1);
-    if (t + 1 < NP) {
-      store_panel(buf ^ 1);
-      if (t + 2 < NP) load_regs((t + 2) * BK);
-    }
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
-      const float a0 = ldsA[buf][wr + fr][kk + fk];
-      const float a1 = ldsA[buf][wr + 32 + fr][kk + fk];
-      const float b0 = ldsB[buf][wc + fr][kk + fk];
-      const float b1 = ldsB[buf][wc + 32 + fr][kk + fk];
+      const float a0 = ldsA[wr + fr][kk + fk];
+      const float a1 = ldsA[wr + 32 + fr][kk + fk];
+      const float b0 = ldsB[wc + fr][kk + fk];
+      const float b1 = ldsB[wc + 32 + fr][kk + fk];
       acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
       acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
       acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
