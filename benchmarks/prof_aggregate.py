@@ -7,7 +7,10 @@ import sqlite3
 import sys
 
 
-def main(db_glob: str, top: int = 40):
+def main(db_glob: str, top: int = 40, tail_frac: float = 0.0):
+    """tail_frac > 0 keeps only dispatches that START in the last fraction
+    of the trace timeline — strips the MIOpen find phase / warmup so the
+    table shows the steady-state round."""
     paths = sorted(glob.glob(db_glob))
     if not paths:
         print(f"no DB matches {db_glob}")
@@ -23,9 +26,14 @@ def main(db_glob: str, top: int = 40):
             continue
         syms = {r[0]: r[1] for r in cur.execute(
             f"SELECT id, display_name FROM {sym_t[0]}")}
+        lo, hi = next(cur.execute(
+            f"SELECT MIN(start), MAX(end) FROM {disp_t[0]}"))
+        cutoff = lo + (hi - lo) * tail_frac
         agg = {}
         for kid, start, end in cur.execute(
                 f"SELECT kernel_id, start, end FROM {disp_t[0]}"):
+            if start < cutoff:
+                continue
             name = syms.get(kid, str(kid))
             ent = agg.setdefault(name, [0, 0.0])
             ent[0] += 1
@@ -42,4 +50,5 @@ def main(db_glob: str, top: int = 40):
 
 if __name__ == "__main__":
     main(sys.argv[1] if len(sys.argv) > 1 else "/tmp/prof2/**/*_results.db",
-         int(sys.argv[2]) if len(sys.argv) > 2 else 40)
+         int(sys.argv[2]) if len(sys.argv) > 2 else 40,
+         float(sys.argv[3]) if len(sys.argv) > 3 else 0.0)
