@@ -173,12 +173,25 @@ class Model(ModelModule):
             chunk = 512
             present = getattr(proto_loader, "present", lambda t: t)
             for i in range(0, protos.shape[0], chunk):
-                data = present(protos[i:i + chunk])
+                rows = protos[i:i + chunk]
+                nrow = rows.shape[0]
+                if nrow < chunk and str(device).startswith("cuda"):
+                    # pad the tail to the FIXED chunk size: a varying tail
+                    # batch is a NEW conv shape every round — MIOpen's FAST
+                    # find was landing it on the naive kernel (10 ms/call,
+                    # 62% of a profiled steady state) — and the padded
+                    # chunk replays the captured graph instead.  Eval-mode
+                    # feature extraction is row-independent; pad rows are
+                    # sliced off below.
+                    rows = torch.cat([rows, protos.new_zeros(
+                        (chunk - nrow, *protos.shape[1:]))])
+                data = present(rows)
                 with autocast(device):
                     sf = self.eval_graphed("herd_fwd", self.head_forward, data,
                                            full=data.shape[0] == chunk)
-                # .float() also COPIES the graph's static output buffer
-                feats.append((sf[1] if isinstance(sf, tuple) else sf).float())
+                feat = sf[1] if isinstance(sf, tuple) else sf
+                # slicing + .float() also COPIES the graph's static buffer
+                feats.append(feat[:nrow].float())
             feats = torch.cat(feats)
         else:
             protos, pids, classes, feats = [], [], [], []
@@ -420,10 +433,17 @@ class Operator(BaseReIDOperator):
                 chunk = 512
                 for i in range(0, data_all.shape[0], chunk):
                     d = data_all[i:i + chunk]
+                    nrow = d.shape[0]
+                    if nrow < chunk:
+                        # fixed-shape tail (see build_examplars): avoids a
+                        # per-round MIOpen find on a fresh conv shape
+                        d = torch.cat([d, data_all.new_zeros(
+                            (chunk - nrow, *data_all.shape[1:]))])
                     with autocast(device):
                         tap = model.eval_graphed(
                             "tap_fwd", lambda t: model.tap_forward(t)[1], d,
                             full=d.shape[0] == chunk)
+                    tap = tap[:nrow]
                     # keep the autocast dtype (fp32 storage would only double
                     # HBM traffic — the tap was COMPUTED in bf16) and store
                     # 4-D taps physically [H, W, C]: the tap is channels-last,
