@@ -36,6 +36,12 @@ os.environ.setdefault("FLREID_GPU_AUGMENT", "1")
 # bf16 wire for the upload gather: halves xGMI traffic (the FedSTIL upload is
 # ~125 MB/client in fp32); the aggregation itself stays fp32 on-device
 os.environ.setdefault("FLREID_COMM_DTYPE", "bf16")
+# FAST-mode find lands fresh shapes on the 1.5-15 ms naive reference convs
+# (profiled at 62% of a steady round before the fixed-shape eval tails);
+# with the naive solvers off it picks a real CK/igemm kernel instead
+os.environ.setdefault("MIOPEN_DEBUG_CONV_DIRECT_NAIVE_CONV_FWD", "0")
+os.environ.setdefault("MIOPEN_DEBUG_CONV_DIRECT_NAIVE_CONV_BWD", "0")
+os.environ.setdefault("MIOPEN_DEBUG_CONV_DIRECT_NAIVE_CONV_WRW", "0")
 
 
 def _seed_miopen_db():
