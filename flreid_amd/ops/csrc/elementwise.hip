@@ -164,6 +164,34 @@ __global__ void compose2_kernel(const Tin* __restrict__ gw,
   // bound on the 13 per-step compositions)
   const int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (i0 >= numel) return;
+  // linear weights (inner==1, atten over the contiguous last dim): the
+  // 4-run reads 4 consecutive atten entries — vectorize those too
+  if (i0 + 4 <= numel && atten && inner == 1 && L % 4 == 0 && L > 1) {
+    const float4 a4 = *(const float4*)(atten + (i0 % L));
+    float4 g, w = {0.f, 0.f, 0.f, 0.f};
+    if constexpr (sizeof(Tin) == 4) {
+      g = *(const float4*)(gw + i0);
+      if (aw) w = *(const float4*)(aw + i0);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        (&g.x)[e] = load_as_float(gw, i0 + e);
+        if (aw) (&w.x)[e] = load_as_float(aw, i0 + e);
+      }
+    }
+    Tout v4[4];
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      float r = fmaf((&a4.x)[e], (&g.x)[e], (&w.x)[e]);
+      if constexpr (sizeof(Tout) == 4) {
+        v4[e] = r;
+      } else {
+        v4[e] = __float2bfloat16(r);
+      }
+    }
+    __builtin_memcpy(out + i0, v4, sizeof(v4));
+    return;
+  }
   if (i0 + 4 <= numel && (L == 1 || !atten || (inner % 4 == 0))) {
     const float a = atten ? atten[(i0 / inner) % L] : 1.0f;
     float4 g, w = {0.f, 0.f, 0.f, 0.f};
