@@ -346,6 +346,11 @@ __global__ __launch_bounds__(256) void transpose_bf16_kernel(
 
 extern "C" void flreid_transpose_bf16(const void* IN, void* OUT, int64_t M,
                                       int64_t N, hipStream_t stream) {
+  if (M % 8 || N % 8) {
+    // the kernel moves 16-B chunks guarded only at their start; a
+    // non-multiple-of-8 edge would overhang into the next row
+    throw std::runtime_error("transpose_bf16: M%8 or N%8 != 0");
+  }
   dim3 grid((unsigned)((M + 63) / 64), (unsigned)((N + 63) / 64));
   hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream,
                      (const __hip_bfloat16*)IN, (__hip_bfloat16*)OUT, M, N);

@@ -40,3 +40,38 @@ def test_async_ckpt_write_then_read(tmp_path, monkeypatch):
     rio.save_ckpt(p, {"w": torch.ones(4)})
     rio.before_ckpt_read()
     assert torch.load(p, weights_only=False)["w"].sum() == 4
+
+
+def test_maybe_sync_batches_distributed_gathers():
+    """maybe_sync gathers only every FLREID_LOG_SYNC_EVERY rounds (and on
+    the final round); single-process mode flushes every call."""
+    import os
+
+    from flreid_amd.runtime.log import ExperimentLog
+
+    class _FakeCtx:
+        is_distributed = True
+
+        def __init__(self):
+            self.gathers = 0
+            self.rank = 0
+
+        def is_rank0(self):
+            return True
+
+        def all_gather_object(self, obj):
+            self.gathers += 1
+            return [obj]
+
+    os.environ["FLREID_LOG_SYNC_EVERY"] = "5"
+    try:
+        log = ExperimentLog("/tmp/flreid_test_log.json")
+        ctx = _FakeCtx()
+        for r in range(1, 13):
+            log.record(f"data.c.{r}.t", {"x": r})
+            log.maybe_sync(ctx, r, 12)
+        # rounds 5, 10 and the final round 12
+        assert ctx.gathers == 3
+        assert log.records["data"]["c"]["12"]["t"]["x"] == 12
+    finally:
+        del os.environ["FLREID_LOG_SYNC_EVERY"]
