@@ -22,7 +22,6 @@ import math
 from typing import Any, Dict, List, Union
 
 import torch
-import torch.nn.functional as F
 from torch import nn
 from torch.utils.data import ConcatDataset, DataLoader
 
@@ -32,7 +31,6 @@ from flreid_amd.methods.common import BaseReIDClient, BaseReIDOperator
 from flreid_amd.modules.model import ModelModule
 from flreid_amd.modules.server import ServerModule
 from flreid_amd.runtime.precision import autocast
-from flreid_amd.tools.utils import get_one_hot
 
 
 class Model(ModelModule):
