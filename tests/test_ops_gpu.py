@@ -731,3 +731,27 @@ def test_patch_merge_ln_bf16():
     expected = torch.nn.functional.layer_norm(
         cat.reshape(b, -1, 4 * c), (4 * c,), gamma, beta)
     assert torch.allclose(out.float(), expected, atol=5e-2, rtol=5e-2)
+
+
+def test_adaptive_conv3x3_production_dispatch():
+    """The production AdaptiveConv2d 3×3 path must take the fused
+    _Conv3x3Fn route (one-pass θ production + policy-routed conv): its
+    output is bf16 channels-last and aw's gradient arrives in fp32 —
+    the eager fallback under autocast would leave a plain autocast chain."""
+    from flreid_amd.models.adaptive import AdaptiveConv2d
+    torch.manual_seed(8)
+    w = (torch.randn(64, 64, 3, 3, device="cuda") * 0.05) \
+        .to(memory_format=torch.channels_last)
+    conv = AdaptiveConv2d(global_weight=w, stride=1, padding=1,
+                          atten_default=0.9).cuda()
+    conv.adaptive_weight.data = conv.adaptive_weight.data \
+        .to(memory_format=torch.channels_last)
+    x = (torch.randn(4, 64, 16, 8, device="cuda").bfloat16()
+         .to(memory_format=torch.channels_last))
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = conv(x)
+    assert y.dtype == torch.bfloat16
+    assert y.is_contiguous(memory_format=torch.channels_last)
+    y.float().sum().backward()
+    assert conv.adaptive_weight.grad is not None
+    assert conv.adaptive_weight.grad.dtype == torch.float32
