@@ -748,10 +748,15 @@ def test_adaptive_conv3x3_production_dispatch():
         .to(memory_format=torch.channels_last)
     x = (torch.randn(4, 64, 16, 8, device="cuda").bfloat16()
          .to(memory_format=torch.channels_last))
+    # the guard must accept exactly these production tensors
+    direct = ops.conv3x3_try(x, conv.global_weight, conv.global_weight_atten,
+                             conv.adaptive_weight)
+    assert direct is not None, "fused conv route declined the production shape"
     with torch.autocast("cuda", dtype=torch.bfloat16):
         y = conv(x)
     assert y.dtype == torch.bfloat16
     assert y.is_contiguous(memory_format=torch.channels_last)
+    assert torch.allclose(y.float(), direct.float(), atol=1e-2)
     y.float().sum().backward()
     assert conv.adaptive_weight.grad is not None
     assert conv.adaptive_weight.grad.dtype == torch.float32
