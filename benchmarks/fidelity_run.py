@@ -42,7 +42,8 @@ def build_cfg(method: str, out_dir: str, args):
         "defaults": {},
     }
     exp = {
-        "exp_name": method, "exp_method": method, "random_seed": 7,
+        "exp_name": method, "exp_method": method,
+        "random_seed": args.seed,
         "exp_opts": {"comm_rounds": args.rounds, "val_interval": args.val_interval,
                      "online_clients": args.clients,
                      "initial_validation": False,
@@ -91,6 +92,7 @@ def main():
     p.add_argument("--ids", type=int, default=16)
     p.add_argument("--imgs", type=int, default=8)
     p.add_argument("--lambda-k", type=int, default=320)
+    p.add_argument("--seed", type=int, default=7)
     p.add_argument("--out", default="gpurun_out/fidelity")
     p.add_argument("--cpu", action="store_true")
     args = p.parse_args()
