@@ -26,7 +26,7 @@
 //
 // dgrad = this same kernel applied to (dy, flipped-transposed θ): dx =
 // conv3x3_s1p1(dy, wT) with wT[c][r][s][k] = w[k][2-r][2-s][c]
-// (conv3x3_wflip below builds wT in one gather pass).
+// (conv3x3_tile mode 1 emits that tile straight from gw/atten/aw).
 //
 // wgrad (conv3x3_wgrad_kernel): per-tap M-reduction GEMM
 // dθ[k,c,(r,s)] = Σ_m dy[m,k]·x[shift_rs(m),c], fp32 accumulate/output,
