@@ -16,8 +16,8 @@
 //    (c>>4)*4 + (c&3)) so one MFMA fragment (the 8 bf16 a lane feeds to
 //    v_mfma_f32_16x16x32_bf16: k = kg+e and 16+kg+e') is 16 CONTIGUOUS
 //    bytes -> one ds_read_b128 per fragment instead of 8 scalar reads;
-//  - cell stride 40 elems (80 B) keeps b128 rows 16B-aligned and spreads
-//    banks;
+//  - cell stride 48 elems (96 B = 24-dword = 8·odd bank stride) keeps
+//    b128 rows 16B-aligned and every read lane group conflict-free;
 //  - T14 pipeline: per c-tile, one barrier; global loads for tile t+2 are
 //    issued before the MFMA burst on tile t (72 MFMAs/wave per barrier);
 //  - grid = (K/64, n_images): with K/64 == 8 the k-block equals the XCD id
@@ -28,9 +28,10 @@
 // conv3x3_s1p1(dy, wT) with wT[c][r][s][k] = w[k][2-r][2-s][c]
 // (conv3x3_tile mode 1 emits that tile straight from gw/atten/aw).
 //
-// wgrad (conv3x3_wgrad_kernel): per-tap M-reduction GEMM
-// dθ[k,c,(r,s)] = Σ_m dy[m,k]·x[shift_rs(m),c], fp32 accumulate/output,
-// border masking applied at the staging loads.
+// wgrad (conv3x3_wgrad_kernel): per-tap M-reduction GEMM on
+// PRE-TRANSPOSED operands, dθ[k,c,(r,s)] = Σ_m dyT[k,m]·xT[c,shift_rs(m)],
+// fp32 accumulate into exclusive split-M partials; border masking applied
+// at the staging loads.
 
 #include "common.h"
 
