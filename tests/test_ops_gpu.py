@@ -760,3 +760,71 @@ def test_adaptive_conv3x3_production_dispatch():
     y.float().sum().backward()
     assert conv.adaptive_weight.grad is not None
     assert conv.adaptive_weight.grad.dtype == torch.float32
+
+
+def test_kd_loss_small_c_and_odd_shapes():
+    """Strided row loops must handle C far below the block size and odd C."""
+    torch.manual_seed(12)
+    for (b, c) in ((3, 7), (5, 300), (1, 1)):
+        zs = torch.randn(b, c, device="cuda", requires_grad=True)
+        zt = torch.randn(b, c, device="cuda")
+        loss = ops.kd_loss(zs, zt, 2.0)
+        loss.backward()
+        zs2 = zs.detach().clone().requires_grad_(True)
+        expected = ref.kd_loss(zs2, zt, 2.0)
+        expected.backward()
+        assert torch.allclose(loss, expected, atol=1e-5, rtol=1e-5), (b, c)
+        assert torch.allclose(zs.grad, zs2.grad, atol=1e-6), (b, c)
+
+
+def test_patch_merge_ln_minimal_resolution():
+    """H = W = 2 (one output token per image)."""
+    torch.manual_seed(13)
+    x = torch.randn(3, 4, 48, device="cuda", requires_grad=True)
+    gamma = torch.randn(192, device="cuda", requires_grad=True)
+    beta = torch.randn(192, device="cuda", requires_grad=True)
+    out = ops.patch_merge_ln(x, gamma, beta, 2, 2)
+    assert out.shape == (3, 1, 192)
+    out.sum().backward()
+    xv = x.detach().view(3, 2, 2, 48)
+    cat = torch.cat([xv[:, 0::2, 0::2], xv[:, 1::2, 0::2],
+                     xv[:, 0::2, 1::2], xv[:, 1::2, 1::2]], dim=-1)
+    expected = torch.nn.functional.layer_norm(
+        cat.reshape(3, 1, 192), (192,), gamma.detach(), beta.detach())
+    assert torch.allclose(out.detach(), expected, atol=1e-4, rtol=1e-4)
+
+
+def test_window_attention_bf16_masked_training():
+    """bf16 + shift mask + autograd in one path (the shifted Swin blocks
+    under autocast)."""
+    torch.manual_seed(14)
+    bw, h, n, d, nw = 8, 3, 49, 32, 4
+    q = torch.randn(bw, h, n, d, device="cuda").bfloat16().requires_grad_(True)
+    k = torch.randn_like(q).requires_grad_(True)
+    v = torch.randn_like(q).requires_grad_(True)
+    bias = torch.randn(h, n, n, device="cuda", requires_grad=True)
+    mask = (torch.randn(nw, n, n, device="cuda") > 0.5).float() * -100.0
+    out = ops.window_attention(q, k, v, bias, mask, d ** -0.5)
+    out.sum().backward()
+    for t, name in ((q, "q"), (k, "k"), (v, "v"), (bias, "bias")):
+        assert t.grad is not None and torch.isfinite(t.grad.float()).all(), name
+
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    bf = bias.detach().clone().requires_grad_(True)
+    ref.window_attention(qf, kf, vf, bf, mask, d ** -0.5, None).sum().backward()
+    assert torch.allclose(q.grad.float(), qf.grad, atol=6e-2, rtol=6e-2)
+    assert torch.allclose(bias.grad, bf.grad, atol=6e-2, rtol=6e-2)
+
+
+def test_compose2_odd_atten_length():
+    """inner==1 with L % 4 != 0 exercises the scalar tail path."""
+    from flreid_amd.ops import compose_theta_bf16, _load_extension
+    ext = _load_extension()
+    gw = torch.randn(10, 6, device="cuda")
+    aw = torch.randn_like(gw)
+    atten = torch.rand(6, device="cuda")
+    out = compose_theta_bf16(ext, gw, atten, aw)
+    expected = (atten * gw + aw).bfloat16()
+    assert torch.allclose(out.float(), expected.float(), atol=2e-2)
