@@ -35,3 +35,20 @@ def test_fedstil_resnet50_round_throughput_floor():
         (f"flagship round throughput {result['value']} img/s fell below the "
          f"{FLOOR_IMGS_PER_SEC} floor (round-1 band 4176-4297; "
          f"ms_per_step={result['ms_per_step']})")
+
+
+@pytest.mark.timeout(600)
+def test_fedstil_swin_round_throughput_floor():
+    """fedstil swin-tiny band measured 2586 (r1) – 2614 (r2) img/s; floor
+    at ~0.75x guards the fused window-attention / K4 path."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"),
+         "--steps", "5", "--warmup", "2", "--model", "swin_transformer_tiny"],
+        capture_output=True, text=True, timeout=540, env=dict(os.environ),
+        cwd=repo)
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert lines, f"no bench JSON line\n{out.stdout[-2000:]}\n{out.stderr[-2000:]}"
+    result = json.loads(lines[-1])
+    assert result["value"] >= 1950.0, \
+        f"swin round throughput {result['value']} img/s below floor"
