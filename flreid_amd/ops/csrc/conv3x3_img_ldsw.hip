@@ -182,6 +182,8 @@ __global__ __launch_bounds__(256, 1) void conv3x3_imgw_fwd_kernel(
       if (ct + 2 < NT) load_x(ct + 2);
     }
     const __hip_bfloat16* wb = lw;
+    __builtin_amdgcn_s_setprio(1);   // keep the MFMA burst ahead of the
+                                     // co-resident staging waves (guide T5)
 #pragma unroll
     for (int r = 0; r < 3; ++r) {
 #pragma unroll
@@ -202,6 +204,7 @@ __global__ __launch_bounds__(256, 1) void conv3x3_imgw_fwd_kernel(
         }
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();           // everyone done reading lw
     if (ct + 1 < NT) {
       store_w();
